@@ -1,0 +1,183 @@
+"""Model unit tests (coverage shape mirrors the reference's
+tests/unit/test_model.py: decorator wiring, tasks, local train/predict,
+save/load round-trips, schedules, artifact resolution)."""
+
+import io
+from pathlib import Path
+
+import pandas as pd
+import pytest
+
+from unionml_amd import Dataset, Model, ModelArtifact, Schedule
+from unionml_amd.exceptions import ModelArtifactNotFound
+from unionml_amd.task import Task
+
+
+def test_decorator_wiring(sklearn_model):
+    assert sklearn_model._trainer is not None
+    assert sklearn_model._predictor is not None
+    assert sklearn_model._evaluator is not None
+    assert sklearn_model._trainer.__unionml_model__ is sklearn_model
+
+
+def test_train_task_interface(sklearn_model):
+    task = sklearn_model.train_task()
+    assert isinstance(task, Task)
+    raw = sklearn_model.dataset.dataset_task()(n=100)
+    model_obj, hp, metrics = task(raw_data=raw, hyperparameters={"max_iter": 200})
+    assert hasattr(model_obj, "predict")
+    assert set(metrics) == {"train", "test"}
+
+
+def test_local_train(sklearn_model):
+    model_obj, metrics = sklearn_model.train(
+        hyperparameters={"max_iter": 200}, n=100
+    )
+    assert sklearn_model.artifact is not None
+    assert sklearn_model.artifact.model_object is model_obj
+    assert 0.0 <= metrics["test"] <= 1.0
+
+
+def test_trainer_kwargs_override(pytorch_model):
+    model_obj, metrics = pytorch_model.train(trainer_kwargs={"epochs": 1}, n=60)
+    assert metrics["train"] >= 0.0
+
+
+def test_predict_equivalence(sklearn_model):
+    sklearn_model.train(hyperparameters={"max_iter": 200}, n=100)
+    features = [{"x1": 0.0, "x2": 0.1, "x3": 0.2}]
+    p1 = sklearn_model.predict(features=features)
+    task = sklearn_model.predict_from_features_task()
+    p2 = task(
+        model_object=sklearn_model.artifact.model_object,
+        features=sklearn_model.dataset.get_features(features),
+    )
+    assert p1 == p2
+
+
+def test_predict_from_reader_kwargs(sklearn_model):
+    sklearn_model.train(hyperparameters={"max_iter": 200}, n=100)
+    preds = sklearn_model.predict(n=10)
+    assert len(preds) == 10
+
+
+def test_predict_without_artifact_raises(sklearn_model):
+    with pytest.raises(ModelArtifactNotFound):
+        sklearn_model.predict(features=[{"x1": 0, "x2": 0, "x3": 0}])
+
+
+def test_save_load_roundtrip_sklearn(tmp_path, sklearn_model):
+    sklearn_model.train(hyperparameters={"max_iter": 200}, n=100)
+    path = tmp_path / "model.joblib"
+    sklearn_model.save(str(path))
+    preds_before = sklearn_model.predict(features=[{"x1": 1, "x2": 2, "x3": 3}])
+
+    fresh = type(sklearn_model)
+    sklearn_model.artifact = None
+    sklearn_model.load(str(path))
+    assert sklearn_model.predict(features=[{"x1": 1, "x2": 2, "x3": 3}]) == preds_before
+
+
+def test_save_load_fileobj(sklearn_model):
+    sklearn_model.train(hyperparameters={"max_iter": 200}, n=100)
+    buf = io.BytesIO()
+    sklearn_model.save(buf)
+    buf.seek(0)
+    sklearn_model.artifact = None
+    sklearn_model.load(buf)
+    assert sklearn_model.artifact is not None
+
+
+def test_save_load_roundtrip_torch(tmp_path, pytorch_model):
+    pytorch_model.train(trainer_kwargs={"epochs": 1}, n=60)
+    path = tmp_path / "model.pt"
+    pytorch_model.save(str(path))
+    features = [{"x1": 0.5, "x2": -0.5, "x3": 0.1}]
+    before = pytorch_model.predict(features=features)
+    pytorch_model.artifact = None
+    pytorch_model.load(str(path))
+    assert pytorch_model.predict(features=features) == before
+
+
+def test_hyperparameter_type_from_init_class(pytorch_model):
+    hp_type = pytorch_model.hyperparameter_type
+    import dataclasses
+
+    assert dataclasses.is_dataclass(hp_type)
+    names = {f.name for f in dataclasses.fields(hp_type)}
+    assert names == {"in_dim", "hidden", "out_dim"}
+
+
+def test_trainer_params(pytorch_model):
+    assert set(pytorch_model.trainer_params) == {"epochs", "lr"}
+
+
+def test_schedule_registration(sklearn_model):
+    sklearn_model.schedule_training("nightly", expression="0 2 * * *")
+    sklearn_model.schedule_prediction(
+        "hourly", expression="0 * * * *", inputs={"n": 5}
+    )
+    assert [s.name for s in sklearn_model.training_schedules] == ["nightly"]
+    assert [s.name for s in sklearn_model.prediction_schedules] == ["hourly"]
+    with pytest.raises(ValueError):
+        sklearn_model.schedule_training("nightly", expression="0 3 * * *")
+    plans = sklearn_model.launchplans()
+    assert {p.name for p in plans} == {"nightly", "hourly"}
+
+
+def test_resolve_model_artifact_precedence(tmp_path, sklearn_model):
+    sklearn_model.train(hyperparameters={"max_iter": 200}, n=100)
+    obj = sklearn_model.artifact.model_object
+    # explicit object wins
+    art = sklearn_model.resolve_model_artifact(model_object=obj)
+    assert art.model_object is obj
+    # file
+    path = tmp_path / "m.joblib"
+    sklearn_model.save(str(path))
+    art2 = sklearn_model.resolve_model_artifact(model_file=str(path))
+    assert hasattr(art2.model_object, "predict")
+    # mutual exclusion
+    with pytest.raises(ValueError):
+        sklearn_model.resolve_model_artifact(model_object=obj, model_file=str(path))
+    # fallback to self.artifact
+    assert sklearn_model.resolve_model_artifact().model_object is obj
+
+
+def test_load_from_env(tmp_path, monkeypatch, sklearn_model):
+    sklearn_model.train(hyperparameters={"max_iter": 200}, n=100)
+    path = tmp_path / "m.joblib"
+    sklearn_model.save(str(path))
+    sklearn_model.artifact = None
+    monkeypatch.setenv("UNIONML_MODEL_PATH", str(path))
+    sklearn_model.load_from_env()
+    assert sklearn_model.artifact is not None
+
+
+def test_prediction_callbacks_swallow_errors():
+    from model_fixtures import make_dataset
+    from sklearn.linear_model import LogisticRegression
+    from typing import List
+
+    calls = []
+
+    def good_cb(model, features, predictions) -> None:
+        calls.append(len(predictions))
+
+    def bad_cb(model, features, predictions) -> None:
+        raise RuntimeError("boom")
+
+    ds = make_dataset()
+    model = Model(name="cb_model", init=LogisticRegression, dataset=ds)
+
+    @model.trainer
+    def trainer(est: LogisticRegression, X: pd.DataFrame, y: pd.DataFrame) -> LogisticRegression:
+        return est.fit(X, y.squeeze())
+
+    @model.predictor(callbacks=[good_cb, bad_cb])
+    def predictor(est: LogisticRegression, X: pd.DataFrame) -> List[float]:
+        return [float(v) for v in est.predict(X)]
+
+    model.train(hyperparameters={"max_iter": 100}, n=80)
+    preds = model.predict(features=[{"x1": 0, "x2": 0, "x3": 0}])
+    assert len(preds) == 1
+    assert calls == [1]  # good callback ran, bad one was swallowed

@@ -1,0 +1,132 @@
+"""Task/Workflow executor + cache + tracker/resolver tests (net-new
+coverage for this build's own L3 layer)."""
+
+import pandas as pd
+import pytest
+
+from unionml_amd.task import Task, Workflow, _TaskCache
+
+
+def test_task_call_and_metadata():
+    t = Task(lambda x: x + 1, "inc")
+    assert t(x=1) == 2
+    assert t.name == "inc"
+
+
+def test_task_cache(tmp_path, monkeypatch):
+    import unionml_amd.task as task_mod
+
+    monkeypatch.setattr(task_mod, "_GLOBAL_TASK_CACHE", _TaskCache(tmp_path))
+    calls = []
+
+    def expensive(n: int):
+        calls.append(n)
+        return n * 2
+
+    t = Task(expensive, "exp", cache=True, cache_version="1")
+    assert t(n=4) == 8
+    assert t(n=4) == 8
+    assert calls == [4]  # second call served from cache
+    t2 = Task(expensive, "exp", cache=True, cache_version="2")
+    assert t2(n=4) == 8
+    assert calls == [4, 4]  # version bump invalidates
+
+
+def test_workflow_dag_execution():
+    wf = Workflow(
+        name="wf",
+        inputs=["a", "b"],
+        outputs=[("sum", ("node", 1, None)), ("prod", ("node", 2, None))],
+    )
+    n0 = wf.add_node(Task(lambda a, b: (a + b, a * b), "both"), {"a": ("input", "a"), "b": ("input", "b")})
+    wf.add_node(Task(lambda x: x, "pick_sum"), {"x": ("node", n0, 0)})
+    wf.add_node(Task(lambda x: x, "pick_prod"), {"x": ("node", n0, 1)})
+    assert wf(a=3, b=4) == (7, 12)
+
+
+def test_workflow_missing_input():
+    wf = Workflow(name="wf", inputs=["a"], outputs=[("out", ("node", 0, None))])
+    wf.add_node(Task(lambda a: a, "id"), {"a": ("input", "a")})
+    with pytest.raises(TypeError):
+        wf()
+
+
+def test_tracker_find_lhs(sklearn_model):
+    # fixtures instantiate in model_fixtures; lhs lookup scans that module
+    import model_fixtures  # noqa: F401
+
+    name = None
+    try:
+        name = sklearn_model.find_lhs()
+    except ValueError:
+        pass  # fixture-local instances are not module-level; that's legal
+    assert name is None or isinstance(name, str)
+
+
+def test_resolver_roundtrip(tmp_path):
+    """Write an app module, resolve its task by loader args."""
+    app = tmp_path / "resolver_app.py"
+    app.write_text(
+        """
+import pandas as pd
+from unionml_amd import Dataset
+
+ds = Dataset(name="resolver_ds", targets=["y"])
+
+@ds.reader
+def reader(n: int = 5) -> pd.DataFrame:
+    return pd.DataFrame({"x": range(n), "y": [i % 2 for i in range(n)]})
+"""
+    )
+    import sys
+
+    sys.path.insert(0, str(tmp_path))
+    try:
+        from unionml_amd.task_resolver import load_task
+
+        task = load_task(
+            ["app-module", "resolver_app", "unionml-obj-name", "ds", "task-name", "dataset_task"]
+        )
+        out = task(n=3)
+        assert len(out) == 3
+        assert task.name == "resolver_ds.reader"
+    finally:
+        sys.path.remove(str(tmp_path))
+        sys.modules.pop("resolver_app", None)
+
+
+def test_loader_args_from_module_level_instance(tmp_path):
+    app = tmp_path / "resolver_app2.py"
+    app.write_text(
+        """
+import pandas as pd
+from unionml_amd import Dataset
+
+ds2 = Dataset(name="resolver_ds2", targets=["y"])
+
+@ds2.reader
+def reader(n: int = 5) -> pd.DataFrame:
+    return pd.DataFrame({"x": range(n), "y": [0] * n})
+"""
+    )
+    import sys
+
+    sys.path.insert(0, str(tmp_path))
+    try:
+        import importlib
+
+        mod = importlib.import_module("resolver_app2")
+        from unionml_amd.task_resolver import loader_args
+
+        args = loader_args(mod.ds2.dataset_task())
+        assert args == [
+            "app-module",
+            "resolver_app2",
+            "unionml-obj-name",
+            "ds2",
+            "task-name",
+            "dataset_task",
+        ]
+    finally:
+        sys.path.remove(str(tmp_path))
+        sys.modules.pop("resolver_app2", None)

@@ -1,0 +1,112 @@
+"""FastAPI serving — attaches ``/``, ``/predict``, ``/health`` routes
+(contract of the reference's unionml/fastapi.py:15-70).
+
+MI355X-native addition: with ``batch=True`` requests are funneled
+through a dynamic micro-batcher whose fused forward is replayed from
+bucketed hipGraph captures (unionml_amd/serving/batcher.py).
+"""
+
+import os
+from typing import Any, Dict, List, Optional, Union
+
+from unionml_amd._logging import logger
+from unionml_amd.artifact import ModelArtifact
+from unionml_amd.exceptions import ModelArtifactNotFound
+
+
+def serving_app(
+    model,
+    app,
+    *,
+    remote: bool = False,
+    app_version: Optional[str] = None,
+    model_version: str = "latest",
+    batch: bool = False,
+    max_batch_size: int = 64,
+    max_delay_ms: float = 2.0,
+):
+    from fastapi import Body, HTTPException
+
+    state = {"batcher": None}
+
+    @app.on_event("startup")
+    async def load_model():
+        # local: $UNIONML_MODEL_PATH; remote: latest successful training
+        # run from the backend registry (reference: fastapi.py:22-34)
+        model_path = os.environ.get("UNIONML_MODEL_PATH")
+        if model_path:
+            model.artifact = ModelArtifact(model.load(model_path))
+        elif remote:
+            model.artifact = model._backend().fetch_model_artifact(
+                model, app_version=app_version, model_version=model_version
+            )
+        if model.artifact is not None and batch:
+            from unionml_amd.serving.batcher import DynamicBatcher
+
+            state["batcher"] = DynamicBatcher(
+                model, max_batch_size=max_batch_size, max_delay_ms=max_delay_ms
+            )
+            state["batcher"].start()
+
+    @app.on_event("shutdown")
+    async def stop_batcher():
+        if state["batcher"] is not None:
+            state["batcher"].stop()
+
+    @app.get("/")
+    async def root():
+        return {
+            "app": "unionml_amd",
+            "model": model.name,
+            "message": f"unionml_amd serving app for model '{model.name}'. "
+            "POST /predict with {'features': [...]} or {'inputs': {...}}.",
+        }
+
+    @app.post("/predict")
+    async def predict(
+        inputs: Optional[Dict[str, Any]] = Body(default=None),
+        features: Optional[Any] = Body(default=None),
+    ):
+        if model.artifact is None:
+            raise HTTPException(status_code=500, detail="model artifact not loaded")
+        if inputs is None and features is None:
+            raise HTTPException(
+                status_code=400, detail="provide one of 'inputs' or 'features'"
+            )
+        try:
+            if features is not None:
+                if state["batcher"] is not None:
+                    return await state["batcher"].submit(features)
+                features = model._dataset.get_features(features)
+                wf = model.predict_from_features_workflow()
+                return _jsonable(
+                    wf(model_object=model.artifact.model_object, features=features)
+                )
+            return _jsonable(model.predict(**(inputs or {})))
+        except HTTPException:
+            raise
+        except Exception as exc:
+            logger.exception("prediction failed")
+            raise HTTPException(status_code=500, detail=str(exc))
+
+    @app.get("/health")
+    async def health():
+        if model.artifact is None:
+            raise HTTPException(status_code=500, detail="model artifact not loaded")
+        return {"status": "ok"}
+
+    return app
+
+
+def _jsonable(predictions):
+    import numpy as np
+
+    if hasattr(predictions, "detach"):  # torch tensor
+        return predictions.detach().cpu().tolist()
+    if isinstance(predictions, np.ndarray):
+        return predictions.tolist()
+    if isinstance(predictions, (list, tuple)):
+        return [_jsonable(p) for p in predictions]
+    if isinstance(predictions, (np.generic,)):
+        return predictions.item()
+    return predictions

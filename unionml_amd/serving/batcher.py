@@ -1,0 +1,169 @@
+"""Dynamic request batcher for online serving.
+
+Coalesces concurrent ``/predict`` requests into one model forward
+(SURVEY.md §2c row 'hipGraph inference step'): requests queue up to
+``max_delay_ms``; the worker drains up to ``max_batch_size`` of them,
+runs ONE prediction over the concatenated features and scatters results
+back to each caller's future.
+
+On an MI355X the per-bucket forward is a hipGraph replay: the model's
+predictor advertises graph capture via
+:class:`unionml_amd.serving.graph_runner.GraphedPredictor` (batch sizes
+are padded up to the nearest bucket {1,2,4,...,max_batch_size} so every
+replayed graph is shape-static). On CPU the same batcher runs the plain
+predictor — one code path, two execution substrates.
+"""
+
+import asyncio
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Any, List, Optional
+
+from unionml_amd._logging import logger
+
+
+@dataclass
+class _Request:
+    features_raw: Any
+    n_rows: int
+    future: "asyncio.Future"
+    loop: "asyncio.AbstractEventLoop"
+
+
+def bucket_for(n: int, max_batch: int) -> int:
+    """Smallest power-of-two bucket >= n (capped at max_batch)."""
+    b = 1
+    while b < n and b < max_batch:
+        b <<= 1
+    return min(b, max_batch)
+
+
+class DynamicBatcher:
+    """Thread-backed micro-batcher feeding a single predictor."""
+
+    def __init__(self, model, max_batch_size: int = 64, max_delay_ms: float = 2.0):
+        self.model = model
+        self.max_batch_size = max_batch_size
+        self.max_delay_s = max_delay_ms / 1000.0
+        self._queue: List[_Request] = []
+        self._cv = threading.Condition()
+        self._stop = False
+        self._thread: Optional[threading.Thread] = None
+        self._graphed = None  # set lazily on first GPU batch
+
+    def start(self):
+        self._thread = threading.Thread(target=self._worker, daemon=True, name="unionml-batcher")
+        self._thread.start()
+
+    def stop(self):
+        with self._cv:
+            self._stop = True
+            self._cv.notify_all()
+        if self._thread is not None:
+            self._thread.join(timeout=5)
+
+    async def submit(self, features_raw: Any):
+        loop = asyncio.get_running_loop()
+        future = loop.create_future()
+        n_rows = len(features_raw) if hasattr(features_raw, "__len__") else 1
+        req = _Request(features_raw=features_raw, n_rows=n_rows, future=future, loop=loop)
+        with self._cv:
+            self._queue.append(req)
+            self._cv.notify()
+        return await future
+
+    # ------------------------------------------------------------------
+
+    def _drain(self) -> List[_Request]:
+        """Collect requests for one batch, waiting up to max_delay for
+        more once the first arrives."""
+        with self._cv:
+            while not self._queue and not self._stop:
+                self._cv.wait(timeout=0.1)
+            if self._stop and not self._queue:
+                return []
+            deadline = time.monotonic() + self.max_delay_s
+            while (
+                sum(r.n_rows for r in self._queue) < self.max_batch_size
+                and time.monotonic() < deadline
+                and not self._stop
+            ):
+                remaining = deadline - time.monotonic()
+                if remaining > 0:
+                    self._cv.wait(timeout=remaining)
+            batch, rows = [], 0
+            while self._queue and rows + self._queue[0].n_rows <= self.max_batch_size:
+                req = self._queue.pop(0)
+                rows += req.n_rows
+                batch.append(req)
+            if not batch and self._queue:  # single oversized request
+                batch.append(self._queue.pop(0))
+            return batch
+
+    def _predict_batch(self, features_list: List[Any]):
+        """One forward over the concatenated features of the batch."""
+        import numpy as np
+        import pandas as pd
+
+        model = self.model
+        ds = model._dataset
+        loaded = [ds.get_features(f) for f in features_list]
+        first = loaded[0]
+        if isinstance(first, pd.DataFrame):
+            merged = pd.concat(loaded, ignore_index=True)
+        elif isinstance(first, np.ndarray):
+            merged = np.concatenate(loaded, axis=0)
+        elif hasattr(first, "shape") and hasattr(first, "device"):  # torch tensor
+            import torch
+
+            merged = torch.cat(loaded, dim=0)
+        else:
+            merged = [row for f in loaded for row in f]
+
+        predictions = self._forward(merged)
+        # scatter back by row counts
+        out = []
+        offset = 0
+        for f in loaded:
+            n = len(f)
+            out.append(predictions[offset : offset + n])
+            offset += n
+        return out
+
+    def _forward(self, merged_features):
+        """The single fused forward. Uses the hipGraph-captured runner
+        when the predictor provides one, else the plain predictor."""
+        model = self.model
+        predictor = model._predictor
+        runner = getattr(predictor, "__unionml_graphed__", None)
+        if runner is not None:
+            if self._graphed is None:
+                from unionml_amd.serving.graph_runner import GraphedPredictor
+
+                self._graphed = GraphedPredictor(
+                    model.artifact.model_object,
+                    runner,
+                    max_batch_size=self.max_batch_size,
+                )
+            return self._graphed(merged_features)
+        return model._run_predictor(model.artifact.model_object, merged_features)
+
+    def _worker(self):
+        from unionml_amd.fastapi import _jsonable
+
+        while True:
+            batch = self._drain()
+            if not batch:
+                if self._stop:
+                    return
+                continue
+            try:
+                results = self._predict_batch([r.features_raw for r in batch])
+                for req, res in zip(batch, results):
+                    req.loop.call_soon_threadsafe(req.future.set_result, _jsonable(res))
+            except Exception as exc:
+                logger.exception("batched prediction failed")
+                for req in batch:
+                    if not req.future.done():
+                        req.loop.call_soon_threadsafe(req.future.set_exception, exc)
