@@ -1,0 +1,121 @@
+// Torch bindings for the CDNA4 tabular hot-path kernels
+// (kernels: tabular_kernels.hip). All launches go to the current torch
+// stream so they compose with torch.cuda.graphs capture and RCCL work.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+extern "C" {
+void launch_standardize_fit(const float*, long long, int, float*, float*, float,
+                            hipStream_t);
+void launch_standardize_apply(const float*, long long, int, const float*,
+                              const float*, unsigned short*, hipStream_t);
+void launch_mlp_step(const unsigned short*, const int*, int, const unsigned short*,
+                     const unsigned short*, const float*, float*, float,
+                     hipStream_t);
+void launch_mlp_predict(const float*, int, const float*, const float*,
+                        const unsigned short*, const unsigned short*, const float*,
+                        int*, float*, hipStream_t);
+void launch_adam_step(float*, unsigned short*, const float*, float*, float*, int*,
+                      float, float, float, float, hipStream_t);
+}
+
+namespace {
+
+hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+void check(const torch::Tensor& t, torch::ScalarType dtype, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a device tensor");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == dtype, name, " has wrong dtype");
+}
+
+const unsigned short* bf16_ptr(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+
+unsigned short* bf16_mut_ptr(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+}  // namespace
+
+void standardize_fit(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd,
+                     double eps) {
+  check(X, torch::kFloat32, "X");
+  check(mean, torch::kFloat32, "mean");
+  check(invstd, torch::kFloat32, "invstd");
+  const int D = (int)X.size(1);
+  TORCH_CHECK(mean.numel() == D && invstd.numel() == D, "mean/invstd size mismatch");
+  launch_standardize_fit(X.data_ptr<float>(), X.size(0), D, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), (float)eps, current_stream());
+}
+
+void standardize_apply(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd,
+                       torch::Tensor out_bf16) {
+  check(X, torch::kFloat32, "X");
+  check(out_bf16, torch::kBFloat16, "out");
+  TORCH_CHECK(out_bf16.numel() == X.numel(), "out size mismatch");
+  launch_standardize_apply(X.data_ptr<float>(), X.size(0), (int)X.size(1),
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           bf16_mut_ptr(out_bf16), current_stream());
+}
+
+void mlp_step(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
+              torch::Tensor W2bf, torch::Tensor master, torch::Tensor grads,
+              double invBtot) {
+  check(Xbf, torch::kBFloat16, "Xbf");
+  check(y, torch::kInt32, "y");
+  check(W1bf, torch::kBFloat16, "W1bf");
+  check(W2bf, torch::kBFloat16, "W2bf");
+  check(master, torch::kFloat32, "master");
+  check(grads, torch::kFloat32, "grads");
+  TORCH_CHECK(Xbf.size(1) == 64, "IN must be 64");
+  TORCH_CHECK(W1bf.numel() == 64 * 32 && W2bf.numel() == 32 * 16, "weight shapes");
+  TORCH_CHECK(grads.numel() >= 2609, "grads must hold 2608 params + loss");
+  launch_mlp_step(bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0),
+                  bf16_ptr(W1bf), bf16_ptr(W2bf), master.data_ptr<float>(),
+                  grads.data_ptr<float>(), (float)invBtot, current_stream());
+}
+
+void mlp_predict(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd,
+                 torch::Tensor W1bf, torch::Tensor W2bf, torch::Tensor master,
+                 torch::Tensor preds, c10::optional<torch::Tensor> probs) {
+  check(X, torch::kFloat32, "X");
+  check(preds, torch::kInt32, "preds");
+  TORCH_CHECK(X.size(1) == 64, "IN must be 64");
+  float* probs_ptr = nullptr;
+  if (probs.has_value()) {
+    check(*probs, torch::kFloat32, "probs");
+    probs_ptr = probs->data_ptr<float>();
+  }
+  launch_mlp_predict(X.data_ptr<float>(), (int)X.size(0), mean.data_ptr<float>(),
+                     invstd.data_ptr<float>(), bf16_ptr(W1bf), bf16_ptr(W2bf),
+                     master.data_ptr<float>(), preds.data_ptr<int>(), probs_ptr,
+                     current_stream());
+}
+
+void adam_step(torch::Tensor master, torch::Tensor bfmirror, torch::Tensor grads,
+               torch::Tensor m, torch::Tensor v, torch::Tensor t_dev, double lr,
+               double beta1, double beta2, double eps) {
+  check(master, torch::kFloat32, "master");
+  check(bfmirror, torch::kBFloat16, "bfmirror");
+  check(grads, torch::kFloat32, "grads");
+  check(m, torch::kFloat32, "m");
+  check(v, torch::kFloat32, "v");
+  check(t_dev, torch::kInt32, "t_dev");
+  launch_adam_step(master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
+                   grads.data_ptr<float>(), m.data_ptr<float>(), v.data_ptr<float>(),
+                   t_dev.data_ptr<int>(), (float)lr, (float)beta1, (float)beta2,
+                   (float)eps, current_stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("standardize_fit", &standardize_fit, "column mean/invstd (CDNA4)");
+  m.def("standardize_apply", &standardize_apply, "(x-mean)*invstd -> bf16 (CDNA4)");
+  m.def("mlp_step", &mlp_step, "fused MLP fwd+bwd step (CDNA4 MFMA)");
+  m.def("mlp_predict", &mlp_predict, "fused standardize+fwd+argmax (CDNA4 MFMA)");
+  m.def("adam_step", &adam_step, "fused Adam on flat master params (CDNA4)");
+}

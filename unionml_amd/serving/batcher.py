@@ -136,16 +136,10 @@ class DynamicBatcher:
         when the predictor provides one, else the plain predictor."""
         model = self.model
         predictor = model._predictor
-        runner = getattr(predictor, "__unionml_graphed__", None)
-        if runner is not None:
+        factory = getattr(predictor, "__unionml_graphed__", None)
+        if factory is not None:
             if self._graphed is None:
-                from unionml_amd.serving.graph_runner import GraphedPredictor
-
-                self._graphed = GraphedPredictor(
-                    model.artifact.model_object,
-                    runner,
-                    max_batch_size=self.max_batch_size,
-                )
+                self._graphed = factory(model.artifact.model_object, self.max_batch_size)
             return self._graphed(merged_features)
         return model._run_predictor(model.artifact.model_object, merged_features)
 
