@@ -38,10 +38,12 @@
 #define WAVES 8
 #define BLOCK (WAVES * 64)
 
-// LDS row strides: +8 bf16 (16 B) pad — see header note.
-#define XS 72            // IN + 8
-#define HS 40            // HID + 8
-#define W2S 24           // CPAD + 8
+// LDS row strides, padded so every ds_read_b128 A-fragment address stays
+// 16 B aligned (row_stride_bytes % 16 == 0 — guide §6 G17) while breaking
+// the power-of-2 bank pattern (guide §6 G4).
+#define XS 72            // IN + 8   (144 B rows)
+#define HS 48            // HID + 16 (96 B rows)
+#define W2S 24           // CPAD + 8 (scalar reads only)
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -414,13 +416,16 @@ mlp_predict_kernel(const float* __restrict__ X,   // [B][IN] raw fp32
         if (ov > best || (ov == best && oc < bcol)) { best = ov; bcol = oc; }
       }
       if (lr == 0 && row0 + row < B) preds[row0 + row] = bcol;
-      if (probs != nullptr && lr < CLS && row0 + row < B) {
-        float m = best;  // row max
-        const float e = __expf(logit - m);
+      if (probs != nullptr) {
+        // shuffle reductions must run with ALL lanes active (an inactive
+        // lane's shfl result is undefined) — only the write is guarded
+        const float e = __expf(logit - best);   // best == row max
         float s = e;
         #pragma unroll
         for (int d = 1; d < 16; d <<= 1) s += __shfl_xor(s, d, 64);
-        probs[(long long)(row0 + row) * CLS + lr] = e / s;
+        if (lr < CLS && row0 + row < B) {
+          probs[(long long)(row0 + row) * CLS + lr] = e / s;
+        }
       }
     }
   }
