@@ -113,6 +113,44 @@ def test_adam_step_vs_reference(ext, dev):
     assert torch.allclose(clf.master.cpu(), master_ref, rtol=1e-4, atol=1e-6)
 
 
+def test_train_steps_kernel_matches_stepwise(ext, dev):
+    """The persistent multi-step kernel must reproduce the per-step
+    (mlp_step + adam_step) sequence exactly (same math, same order)."""
+    from unionml_amd.ops import reference as ref
+    from unionml_amd.ops.tabular import ADAM_BETA1, ADAM_BETA2, ADAM_EPS, TabularMLP
+
+    torch.manual_seed(11)
+    N, B, n_steps = 512, 128, 9
+    Xbf = (torch.randn(N, 64) * 1.1).bfloat16().to(dev)
+    y = torch.randint(0, 10, (N,), dtype=torch.int32, device=dev)
+
+    # per-step path
+    a = TabularMLP(device=dev, seed=4)
+    for s in range(n_steps):
+        off = (s % (N // B)) * B
+        a.grads.zero_()
+        ext.mlp_step(Xbf[off : off + B], y[off : off + B], a.W1bf, a.W2bf,
+                     a.master, a.grads, 1.0 / B)
+        ext.adam_step(a.master, a.bfmirror, a.grads, a.m, a.v, a.t_dev,
+                      1e-3, ADAM_BETA1, ADAM_BETA2, ADAM_EPS)
+    torch.cuda.synchronize()
+
+    # persistent kernel
+    b = TabularMLP(device=dev, seed=4)
+    loss_out = b.grads[ref.NPARAM : ref.NPARAM + 1]
+    ok = ext.mlp_train_steps(Xbf, y, B, n_steps, b.master, b.bfmirror,
+                             b.m, b.v, b.t_dev, loss_out,
+                             1e-3, ADAM_BETA1, ADAM_BETA2, ADAM_EPS)
+    assert ok
+    torch.cuda.synchronize()
+
+    assert int(b.t_dev.item()) == n_steps
+    err = (a.master - b.master).abs().max().item()
+    assert err < 1e-5, f"master mismatch {err}"
+    err_m = (a.m - b.m).abs().max().item()
+    assert err_m < 1e-6, f"moment mismatch {err_m}"
+
+
 def test_train_digits_gpu_accuracy(ext, dev):
     from sklearn.datasets import load_digits
 

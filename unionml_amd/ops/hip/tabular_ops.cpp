@@ -13,6 +13,9 @@ void launch_standardize_apply(const float*, long long, int, const float*,
 void launch_mlp_step(const unsigned short*, const int*, int, const unsigned short*,
                      const unsigned short*, const float*, float*, float,
                      hipStream_t);
+int launch_mlp_train_steps(const unsigned short*, const int*, long long, int, int,
+                           float*, unsigned short*, float*, float*, int*, float*,
+                           float, float, float, float, hipStream_t);
 void launch_mlp_predict(const float*, int, const float*, const float*,
                         const unsigned short*, const unsigned short*, const float*,
                         int*, float*, hipStream_t);
@@ -80,6 +83,29 @@ void mlp_step(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
                   grads.data_ptr<float>(), (float)invBtot, current_stream());
 }
 
+bool mlp_train_steps(torch::Tensor Xbf, torch::Tensor y, int64_t batch,
+                     int64_t n_steps, torch::Tensor master, torch::Tensor bfmirror,
+                     torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
+                     torch::Tensor loss_out, double lr, double beta1, double beta2,
+                     double eps) {
+  check(Xbf, torch::kBFloat16, "Xbf");
+  check(y, torch::kInt32, "y");
+  check(master, torch::kFloat32, "master");
+  check(bfmirror, torch::kBFloat16, "bfmirror");
+  check(m, torch::kFloat32, "m");
+  check(v, torch::kFloat32, "v");
+  check(t_dev, torch::kInt32, "t_dev");
+  check(loss_out, torch::kFloat32, "loss_out");
+  TORCH_CHECK(Xbf.size(1) == 64, "IN must be 64");
+  const int rc = launch_mlp_train_steps(
+      bf16_ptr(Xbf), y.data_ptr<int>(), Xbf.size(0), (int)batch, (int)n_steps,
+      master.data_ptr<float>(), bf16_mut_ptr(bfmirror), m.data_ptr<float>(),
+      v.data_ptr<float>(), t_dev.data_ptr<int>(), loss_out.data_ptr<float>(),
+      (float)lr, (float)beta1, (float)beta2, (float)eps, current_stream());
+  TORCH_CHECK(rc != -2, "mlp_train_steps: hipFuncSetAttribute(LDS) failed");
+  return rc == 0;   // false -> shape constraints unmet, caller falls back
+}
+
 void mlp_predict(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd,
                  torch::Tensor W1bf, torch::Tensor W2bf, torch::Tensor master,
                  torch::Tensor preds, c10::optional<torch::Tensor> probs) {
@@ -116,6 +142,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("standardize_fit", &standardize_fit, "column mean/invstd (CDNA4)");
   m.def("standardize_apply", &standardize_apply, "(x-mean)*invstd -> bf16 (CDNA4)");
   m.def("mlp_step", &mlp_step, "fused MLP fwd+bwd step (CDNA4 MFMA)");
+  m.def("mlp_train_steps", &mlp_train_steps,
+        "persistent multi-step training kernel (weights+Adam resident in LDS)");
   m.def("mlp_predict", &mlp_predict, "fused standardize+fwd+argmax (CDNA4 MFMA)");
   m.def("adam_step", &adam_step, "fused Adam on flat master params (CDNA4)");
 }

@@ -134,6 +134,26 @@ class TabularMLP:
         ]
         allreduce = world_size > 1
 
+        # single-GPU flagship: the persistent multi-step kernel (weights +
+        # Adam state resident in LDS, zero launches per step). Requires
+        # batch % 128 == 0 and n % batch == 0; otherwise fall through.
+        if (
+            self.use_hip
+            and not allreduce
+            and batch_size % 128 == 0
+            and n % batch_size == 0
+        ):
+            loss_out = self.grads[NPARAM : NPARAM + 1]
+            n_steps = epochs * (n // batch_size)
+            ok = hip_ext().mlp_train_steps(
+                Xbf, y, batch_size, n_steps, self.master, self.bfmirror,
+                self.m, self.v, self.t_dev, loss_out,
+                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+            )
+            if ok:
+                torch.cuda.synchronize(self.device)
+                return float(loss_out.item())
+
         def run_epoch():
             last = None
             for off, bs in batches:
