@@ -1,39 +1,49 @@
 // CDNA4 (gfx950 / MI355X) kernels for the default tabular hot path.
 //
-// Implements the kernel set of SURVEY.md §2c (net-new designs; the
-// reference unionai-oss/unionml is pure Python and has no kernels):
+// Kernel set of SURVEY.md §2c (net-new designs; the reference
+// unionai-oss/unionml is pure Python and has no kernels):
 //   - standardize_fit / standardize_apply : per-column (x-mean)*invstd, fp32 -> bf16
 //   - mlp_step        : fused fwd+bwd of the digits MLP
-//                       (IN=64 -> HID=32 relu -> CLS=10 softmax/xent)
-//                       producing fp32 grads + loss in ONE launch (DP path:
-//                       an RCCL all-reduce of the 2.6k-float grad buffer sits
-//                       between this and adam_step)
-//   - adam_step       : single-block fused Adam on the flat fp32 master,
-//                       emitting the bf16 compute mirror
-//   - mlp_train_steps : the single-GPU flagship — a persistent single-
-//                       workgroup kernel running N optimizer steps in ONE
-//                       launch: bf16 weights, fp32 master and Adam moments
-//                       all live in LDS across steps; only the minibatch
-//                       rows stream from HBM. Removes every per-step launch,
-//                       zeroing pass, and HBM weight round-trip.
-//   - mlp_predict     : fused standardize + fwd + argmax (serving hot path,
-//                       hipGraph-captured per batch bucket)
+//                       (IN=64 -> HID=32 relu -> CLS=10 softmax/xent) producing
+//                       fp32 grads + loss in ONE launch (DP path: the RCCL
+//                       all-reduce of the 2.6k-float grad buffer sits between
+//                       this and adam_step)
+//   - adam_step       : single-block fused Adam on the flat fp32 master
+//   - mlp_train_steps : single-GPU flagship — a persistent single-workgroup
+//                       kernel running N optimizer steps in ONE launch with
+//                       bf16 weights, fp32 master and Adam moments resident
+//                       in LDS across steps; only minibatch rows stream from
+//                       HBM. Zero per-step launches.
+//   - mlp_predict     : fused standardize + fwd + argmax (serving hot path)
 //
-// Design notes (see /opt/skills/guides/cdna_hip_programming.md):
-//   * wave64; training blocks = 8 waves (512 threads); each wave owns 16
-//     batch rows per 128-row chunk.
-//   * GEMM-shaped work on MFMA: __builtin_amdgcn_mfma_f32_16x16x32_bf16
-//     (gfx950 2xK form), fp32 accumulate. Fragment mapping (verified on
-//     hardware by tests/test_gpu_kernels.py with asymmetric operands):
-//       A[16x32]:  lane l holds A[l&15][(l>>4)*8 + i], i = 0..7
-//       B[32x16]:  lane l holds B[(l>>4)*8 + i][l&15]
-//       C/D[16x16]: lane l, reg r holds D[(l>>4)*4 + r][l&15]
-//   * LDS rows padded to strides whose ds_read_b128 bank slots are all
-//     distinct (stride 144 B for X, 80 B for H/dL/dH) and 16 B aligned
-//     (guide §6 G4/G17).
-//   * Every launch is stream-ordered and hipGraph-capturable (no mallocs,
-//     no syncs — guide Guideline 9). Step counters live in device memory
-//     so Adam bias correction stays exact under graph replay.
+// Fragment-oriented design (2nd iteration, after PMC analysis — the first
+// scalar-read version spent 59 % of wave cycles parked on LDS latency):
+// every MFMA operand fragment is ONE ds_read_b128. An MFMA B-fragment needs
+// its K axis contiguous, so each GEMM is oriented (plain or transposed
+// output) such that both operands have a row-major LDS image whose rows run
+// along K; the small intermediates (H, dLogits) are written in BOTH
+// orientations (dual scalar stores are far cheaper than strided scalar
+// loads + register packing on the consume side).
+//
+//   H^T  = W1T @ B(Xs)        A=W1T[h][k_in]   B-img = Xs[row][in]
+//   L^T  = W2T @ B(Hs)        A=W2T[c][k_h]    B-img = Hs[row][h]
+//   dH^T = W2s @ B(DLs)       A=W2s[h][k_c]    B-img = DLs[row][c]
+//   dW1^T= DHT @ B(XT)        A=DHT[h][k_row]  B-img = XT[in][row]
+//   dW2  = HT  @ B(DLT)       A=HT[h][k_row]   B-img = DLT[c][row]
+//
+// MFMA: __builtin_amdgcn_mfma_f32_16x16x32_bf16 (gfx950 2xK form), fp32
+// accumulate. Fragment mapping (verified on hardware by
+// tests/test_gpu_kernels.py with asymmetric operands):
+//   A[16x32]:  lane l holds A[l&15][(l>>4)*8 + i], i = 0..7
+//   B[32x16]:  lane l holds B[(l>>4)*8 + i][l&15]
+//   C/D[16x16]: lane l, reg r holds D[(l>>4)*4 + r][l&15]
+//
+// LDS row strides are multiples of 16 B (b128 alignment, guide §6 G17) with
+// bank-slot-distinct row offsets (guide §6 G4). No fp32 division in device
+// code (v_div_scale sequences dominated the first version) — v_rcp instead.
+// All launches are stream-ordered and hipGraph-capturable (guide G9); step
+// counters live in device memory so Adam bias correction is exact under
+// graph replay.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -46,11 +56,11 @@
 #define WAVES 8
 #define BLOCK (WAVES * 64)
 
-// LDS row strides (in bf16 elements). Both keep every b128 A-fragment
-// address 16 B aligned and give 16 distinct bank slots over 16 rows.
-#define XS 72            // IN + 8   -> 144 B rows
-#define HS 40            // HID + 8  -> 80 B rows
-#define W2S 24           // CPAD + 8 (scalar reads only)
+// LDS strides in bf16 elements (row-major [rows][stride]):
+#define XS 72            // Xs  [ROWS][72]  (144 B rows)
+#define HS 40            // Hs/DLs [ROWS][40] (80 B rows)
+#define TS 136           // XT/HT/DLT/DHT [*][136] (272 B rows)
+#define WS 40            // W1T/W2s/W2T [*][40] (80 B rows; K-padded zeros)
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -70,6 +80,10 @@ __device__ __forceinline__ u16 f2bf(float f) {
   return (u16)(c.u >> 16);
 }
 
+__device__ __forceinline__ float fast_rcp(float x) {
+  return __builtin_amdgcn_rcpf(x);   // v_rcp_f32, ~1 ulp — no div sequence
+}
+
 // ---------------------------------------------------------------------------
 // standardize
 // ---------------------------------------------------------------------------
@@ -78,7 +92,7 @@ extern "C" __global__ void __launch_bounds__(256)
 standardize_fit_kernel(const float* __restrict__ X, long long N, int D,
                        float* __restrict__ mean, float* __restrict__ invstd,
                        float eps) {
-  const int col = blockIdx.x;            // one workgroup per column
+  const int col = blockIdx.x;
   if (col >= D) return;
   double s = 0.0, s2 = 0.0;
   for (long long r = threadIdx.x; r < N; r += blockDim.x) {
@@ -134,156 +148,246 @@ standardize_apply_kernel(const float* __restrict__ X, long long n_elems, int D,
 #define NPARAM 2608
 
 // ---------------------------------------------------------------------------
-// shared device math for one 128-row chunk. All pointers are LDS arrays with
-// the strides above. Each wave owns rows [wave*16, wave*16+16).
-// Returns nothing; writes Hs/DLs/DHs, accumulates db/loss in LDS and the
-// caller's dW accumulators in registers.
+// dynamic-LDS carve shared by mlp_step / mlp_train_steps
 // ---------------------------------------------------------------------------
 
-struct ChunkAcc {
-  f32x4 dW1;        // this wave's dW1 tile (mt = wave>>1, nt = wave&1)
-  f32x4 dW2;        // waves 0-1: dW2 tile (mt2 = wave)
+#define ALIGN16(x) (((x) + 15) & ~15)
+#define C_XS   0
+#define C_XT   ALIGN16(C_XS  + ROWS * XS * 2)
+#define C_HS   ALIGN16(C_XT  + IN * TS * 2)
+#define C_HT   ALIGN16(C_HS  + ROWS * HS * 2)
+#define C_DLS  ALIGN16(C_HT  + HID * TS * 2)
+#define C_DLT  ALIGN16(C_DLS + ROWS * HS * 2)
+#define C_DHT  ALIGN16(C_DLT + CPAD * TS * 2)
+#define C_W1T  ALIGN16(C_DHT + HID * TS * 2)
+#define C_W2S  ALIGN16(C_W1T + HID * XS * 2)
+#define C_W2T  ALIGN16(C_W2S + HID * WS * 2)
+#define C_DB1  ALIGN16(C_W2T + CPAD * WS * 2)
+#define C_DB2  ALIGN16(C_DB1 + HID * 4)
+#define C_LOSS ALIGN16(C_DB2 + CPAD * 4)
+#define C_IMG_TOTAL ALIGN16(C_LOSS + 16)
+// steps kernel appends optimizer state after the images:
+#define C_MASTER C_IMG_TOTAL
+#define C_M    ALIGN16(C_MASTER + NPARAM * 4)
+#define C_V    ALIGN16(C_M + NPARAM * 4)
+#define C_STEPS_TOTAL ALIGN16(C_V + NPARAM * 4)
+
+struct Lds {
+  u16 (*Xs)[XS];
+  u16 (*XT)[TS];
+  u16 (*Hs)[HS];
+  u16 (*HT)[TS];
+  u16 (*DLs)[HS];
+  u16 (*DLT)[TS];
+  u16 (*DHT)[TS];
+  u16 (*W1T)[XS];
+  u16 (*W2s)[WS];
+  u16 (*W2T)[WS];
+  float* db1;
+  float* db2;
+  float* loss;
 };
 
+__device__ __forceinline__ Lds carve(char* smem) {
+  Lds L;
+  L.Xs = (u16(*)[XS])(smem + C_XS);
+  L.XT = (u16(*)[TS])(smem + C_XT);
+  L.Hs = (u16(*)[HS])(smem + C_HS);
+  L.HT = (u16(*)[TS])(smem + C_HT);
+  L.DLs = (u16(*)[HS])(smem + C_DLS);
+  L.DLT = (u16(*)[TS])(smem + C_DLT);
+  L.DHT = (u16(*)[TS])(smem + C_DHT);
+  L.W1T = (u16(*)[XS])(smem + C_W1T);
+  L.W2s = (u16(*)[WS])(smem + C_W2S);
+  L.W2T = (u16(*)[WS])(smem + C_W2T);
+  L.db1 = (float*)(smem + C_DB1);
+  L.db2 = (float*)(smem + C_DB2);
+  L.loss = (float*)(smem + C_LOSS);
+  return L;
+}
+
+// fill weight images (W1T/W2s/W2T incl. zero K-pads) from bf16 weight arrays
+__device__ __forceinline__ void load_weight_images(const Lds& L,
+                                                   const u16* __restrict__ W1bf,
+                                                   const u16* __restrict__ W2bf) {
+  const int tid = threadIdx.x;
+  for (int i = tid; i < HID * IN; i += BLOCK) {       // W1T[h][k] = W1[k][h]
+    const int h = i / IN, k = i % IN;
+    L.W1T[h][k] = W1bf[k * HID + h];
+  }
+  for (int i = tid; i < HID * 32; i += BLOCK) {       // W2s[h][c], c-pad 16..31 = 0
+    const int h = i / 32, c = i % 32;
+    L.W2s[h][c] = (c < CPAD) ? W2bf[h * CPAD + c] : (u16)0;
+  }
+  for (int i = tid; i < CPAD * HID; i += BLOCK) {     // W2T[c][h] = W2[h][c]
+    const int c = i / HID, h = i % HID;
+    L.W2T[c][h] = W2bf[h * CPAD + c];
+  }
+}
+
+// cooperative X-chunk load: Xs row-major + XT transposed, zero batch tail
+__device__ __forceinline__ void load_x_chunk(const u16* __restrict__ Xbf,
+                                             const Lds& L, long long row0,
+                                             long long Nvalid) {
+  for (int i = threadIdx.x; i < ROWS * (IN / 8); i += BLOCK) {
+    const int r = i / (IN / 8);
+    const int c = (i % (IN / 8)) * 8;
+    bf16x8 v;
+    if (row0 + r < Nvalid) {
+      v = *(const bf16x8*)&Xbf[(row0 + r) * IN + c];
+    } else {
+      v = (bf16x8){0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    *(bf16x8*)&L.Xs[r][c] = v;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) L.XT[c + j][r] = (u16)v[j];
+  }
+}
+
+// zero DLs K-pad cols [CPAD,32) once per launch (never rewritten)
+__device__ __forceinline__ void zero_dl_pad(const Lds& L) {
+  for (int i = threadIdx.x; i < ROWS * CPAD; i += BLOCK) {
+    const int r = i / CPAD, c = CPAD + (i % CPAD);
+    L.DLs[r][c] = 0;
+  }
+}
+
+struct ChunkAcc {
+  f32x4 dW1;        // wave's dW1^T tile: h-tile = wave&1, in-tile = wave>>1
+  f32x4 dW2;        // waves 0-1: dW2 tile (h-tile = wave)
+};
+
+// one 128-row chunk: fwd + bwd, accumulating weight grads in acc_io and
+// bias grads / loss in LDS. b1/b2 point at fp32 biases (any addr space).
 __device__ __forceinline__ void chunk_fwd_bwd(
-    const u16 (*Xs)[XS], u16 (*Hs)[HS], u16 (*DLs)[HS], u16 (*DHs)[HS],
-    const u16 (*W1s)[HS], const u16 (*W2s)[W2S],
-    const float* b1, const float* b2,
-    const int* __restrict__ y, int row0, int B, float invBtot,
-    float* db1_s, float* db2_s, float* loss_s, ChunkAcc& acc_io) {
+    const Lds& L, const float* b1, const float* b2,
+    const int* __restrict__ y, long long row0, long long Nvalid, float invBtot,
+    ChunkAcc& acc_io) {
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int l = tid & 63;
   const int lg = l >> 4, lr = l & 15;
-  const int wrow = wave * 16;
+  const int wrow = wave * 16;                       // this wave's 16-row block
 
-  // fwd: H = relu(X @ W1 + b1)
-  for (int nt = 0; nt < HID / 16; ++nt) {
+  // ---- fwd1: H^T = W1T @ B(Xs);  D: h = mt*16+lg*4+r, row = wrow+lr --------
+  #pragma unroll
+  for (int mt = 0; mt < HID / 16; ++mt) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
     for (int ks = 0; ks < IN / 32; ++ks) {
-      const bf16x8 a = *(const bf16x8*)&Xs[wrow + lr][ks * 32 + lg * 8];
-      bf16x8 b;
-      #pragma unroll
-      for (int i = 0; i < 8; ++i) b[i] = (short)W1s[ks * 32 + lg * 8 + i][nt * 16 + lr];
+      const bf16x8 a = *(const bf16x8*)&L.W1T[mt * 16 + lr][ks * 32 + lg * 8];
+      const bf16x8 b = *(const bf16x8*)&L.Xs[wrow + lr][ks * 32 + lg * 8];
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float h = acc[r] + b1[nt * 16 + lr];
-      h = h > 0.f ? h : 0.f;
-      Hs[wrow + lg * 4 + r][nt * 16 + lr] = f2bf(h);
+      const int h = mt * 16 + lg * 4 + r;
+      const int row = wrow + lr;
+      float hv = acc[r] + b1[h];
+      hv = hv > 0.f ? hv : 0.f;
+      const u16 hb = f2bf(hv);
+      L.Hs[row][h] = hb;
+      L.HT[h][row] = hb;
     }
   }
-  __syncthreads();   // Hs complete (bwd-W reads cross-wave rows)
+  __syncthreads();   // Hs/HT complete
 
-  // logits + softmax + dlogits (intra-wave: own 16 rows)
+  // ---- fwd2: L^T = W2T @ B(Hs);  D: c = lg*4+r, row = wrow+lr --------------
   {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    const bf16x8 a = *(const bf16x8*)&Hs[wrow + lr][lg * 8];
-    bf16x8 b;
-    #pragma unroll
-    for (int i = 0; i < 8; ++i) b[i] = (short)W2s[lg * 8 + i][lr];
+    const bf16x8 a = *(const bf16x8*)&L.W2T[lr][lg * 8];
+    const bf16x8 b = *(const bf16x8*)&L.Hs[wrow + lr][lg * 8];
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
 
+    const int row = wrow + lr;
+    const bool valid = (row0 + row) < Nvalid;
+    const int label = valid ? y[row0 + row] : -1;
+
+    float logit[4];
+    float m = -1e30f;
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wrow + lg * 4 + r;
-      const bool valid_row = (row0 + row) < B;
-      float logit = acc[r] + b2[lr];
-      if (lr >= CLS) logit = -1e30f;
-      float m = logit;
-      #pragma unroll
-      for (int d = 1; d < 16; d <<= 1) m = fmaxf(m, __shfl_xor(m, d, 64));
-      const float e = (lr < CLS) ? __expf(logit - m) : 0.f;
-      float s = e;
-      #pragma unroll
-      for (int d = 1; d < 16; d <<= 1) s += __shfl_xor(s, d, 64);
-      const int label = valid_row ? y[row0 + row] : -1;
-      const float dl = valid_row ? (e / s - (lr == label ? 1.f : 0.f)) * invBtot : 0.f;
-      DLs[row][lr] = f2bf(dl);
-      if (valid_row && lr == label) {
-        atomicAdd(loss_s, -(logit - m - __logf(s)) * invBtot);
+      const int c = lg * 4 + r;
+      logit[r] = (c < CLS) ? acc[r] + b2[c] : -1e30f;
+      m = fmaxf(m, logit[r]);
+    }
+    // row max / sum across the 4 lane-groups holding this row's classes
+    m = fmaxf(m, __shfl_xor(m, 16, 64));
+    m = fmaxf(m, __shfl_xor(m, 32, 64));
+    float e[4], s = 0.f;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int c = lg * 4 + r;
+      e[r] = (c < CLS) ? __expf(logit[r] - m) : 0.f;
+      s += e[r];
+    }
+    s += __shfl_xor(s, 16, 64);
+    s += __shfl_xor(s, 32, 64);
+    const float rs = fast_rcp(s);
+    const float logs = __logf(s);
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int c = lg * 4 + r;
+      const float dl = valid ? (e[r] * rs - (c == label ? 1.f : 0.f)) * invBtot : 0.f;
+      const u16 dlb = f2bf(dl);
+      L.DLs[row][c] = dlb;
+      L.DLT[c][row] = dlb;
+      if (valid && c == label) {
+        atomicAdd(L.loss, -(logit[r] - m - logs) * invBtot);
       }
-      atomicAdd(&db2_s[lr], dl);
+      atomicAdd(&L.db2[c], dl);
     }
   }
+  __syncthreads();   // DLs/DLT complete
 
-  // dH = dlogits @ W2^T (B-operand read transposed from W2s; k >= CPAD is 0)
-  for (int nt = 0; nt < HID / 16; ++nt) {
+  // ---- dH^T = W2s @ B(DLs);  D: h = mt*16+lg*4+r, row = wrow+lr ------------
+  #pragma unroll
+  for (int mt = 0; mt < HID / 16; ++mt) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    const bf16x8 a = *(const bf16x8*)&DLs[wrow + lr][lg * 8];
-    bf16x8 b;
-    #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      const int c = lg * 8 + i;
-      b[i] = (c < CPAD) ? (short)W2s[nt * 16 + lr][c] : (short)0;
-    }
+    const bf16x8 a = *(const bf16x8*)&L.W2s[mt * 16 + lr][lg * 8];
+    const bf16x8 b = *(const bf16x8*)&L.DLs[wrow + lr][lg * 8];
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wrow + lg * 4 + r;
-      const float h = bf2f(Hs[row][nt * 16 + lr]);
-      const float dh = h > 0.f ? acc[r] : 0.f;
-      DHs[row][nt * 16 + lr] = f2bf(dh);
-      atomicAdd(&db1_s[nt * 16 + lr], dh);
+      const int h = mt * 16 + lg * 4 + r;
+      const int row = wrow + lr;
+      const float hv = bf2f(L.HT[h][row]);
+      float dh = hv > 0.f ? acc[r] : 0.f;
+      L.DHT[h][row] = f2bf(dh);
+      // db1[h]: sum this row-block's contribution across the 16 lr lanes
+      dh += __shfl_xor(dh, 1, 64);
+      dh += __shfl_xor(dh, 2, 64);
+      dh += __shfl_xor(dh, 4, 64);
+      dh += __shfl_xor(dh, 8, 64);
+      if (lr == 0) atomicAdd(&L.db1[h], dh);
     }
   }
-  __syncthreads();   // DHs/DLs complete for cross-wave bwd-W reads
+  __syncthreads();   // DHT complete
 
-  // dW1 = X^T @ dH: wave w owns tile (mt = w>>1, nt = w&1)
+  // ---- dW1^T += DHT @ B(XT): wave tile (ht = w&1, it = w>>1) ---------------
   {
-    const int mt = wave >> 1, nt = wave & 1;
+    const int ht = wave & 1, it = wave >> 1;
     f32x4 acc = acc_io.dW1;
+    #pragma unroll
     for (int ks = 0; ks < ROWS / 32; ++ks) {
-      bf16x8 a, b;
-      #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        const int k = ks * 32 + lg * 8 + i;        // batch row
-        a[i] = (short)Xs[k][mt * 16 + lr];          // A[m][k] = X[k][m]
-        b[i] = (short)DHs[k][nt * 16 + lr];
-      }
+      const bf16x8 a = *(const bf16x8*)&L.DHT[ht * 16 + lr][ks * 32 + lg * 8];
+      const bf16x8 b = *(const bf16x8*)&L.XT[it * 16 + lr][ks * 32 + lg * 8];
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
     acc_io.dW1 = acc;
   }
-  // dW2 = H^T @ dL: waves 0-1
+  // ---- dW2 += HT @ B(DLT): waves 0-1 (h-tile = wave) -----------------------
   if (wave < 2) {
     f32x4 acc = acc_io.dW2;
+    #pragma unroll
     for (int ks = 0; ks < ROWS / 32; ++ks) {
-      bf16x8 a, b;
-      #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        const int k = ks * 32 + lg * 8 + i;
-        a[i] = (short)Hs[k][wave * 16 + lr];
-        b[i] = (short)DLs[k][lr];
-      }
+      const bf16x8 a = *(const bf16x8*)&L.HT[wave * 16 + lr][ks * 32 + lg * 8];
+      const bf16x8 b = *(const bf16x8*)&L.DLT[lr][ks * 32 + lg * 8];
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
     acc_io.dW2 = acc;
   }
   __syncthreads();   // chunk arrays free for reuse
-}
-
-// cooperative X-chunk load (global bf16 -> LDS, zero-padded batch tail)
-__device__ __forceinline__ void load_x_chunk(const u16* __restrict__ Xbf,
-                                             u16 (*Xs)[XS], int row0, int B) {
-  for (int i = threadIdx.x; i < ROWS * (IN / 8); i += BLOCK) {
-    const int r = i / (IN / 8);
-    const int c = (i % (IN / 8)) * 8;
-    if (row0 + r < B) {
-      *(bf16x8*)&Xs[r][c] = *(const bf16x8*)&Xbf[(long long)(row0 + r) * IN + c];
-    } else {
-      for (int k = 0; k < 8; ++k) Xs[r][c + k] = 0;
-    }
-  }
-}
-
-// zero DLs K-pad columns [CPAD, 32) once (rows never rewritten there)
-__device__ __forceinline__ void zero_dl_pad(u16 (*DLs)[HS]) {
-  for (int i = threadIdx.x; i < ROWS * CPAD; i += BLOCK) {
-    const int r = i / CPAD, c = CPAD + (i % CPAD);
-    DLs[r][c] = 0;
-  }
 }
 
 // ---------------------------------------------------------------------------
@@ -295,47 +399,35 @@ mlp_step_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y, int B,
                 const u16* __restrict__ W1bf, const u16* __restrict__ W2bf,
                 const float* __restrict__ master,
                 float* __restrict__ grads, float invBtot) {
-  __shared__ u16 Xs[ROWS][XS];
-  __shared__ u16 Hs[ROWS][HS];
-  __shared__ u16 DLs[ROWS][HS];
-  __shared__ u16 DHs[ROWS][HS];
-  __shared__ u16 W1s[IN][HS];
-  __shared__ u16 W2s[HID][W2S];
-  __shared__ float db1_s[HID], db2_s[CPAD], loss_s;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const Lds L = carve(smem);
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int l = tid & 63;
   const int lg = l >> 4, lr = l & 15;
-  const int row0 = blockIdx.x * ROWS;
+  const long long row0 = (long long)blockIdx.x * ROWS;
 
-  if (tid < HID) db1_s[tid] = 0.f;
-  if (tid < CPAD) db2_s[tid] = 0.f;
-  if (tid == 0) loss_s = 0.f;
-  zero_dl_pad(DLs);
-  load_x_chunk(Xbf, Xs, row0, B);
-  for (int i = tid; i < IN * (HID / 8); i += BLOCK) {
-    const int r = i / (HID / 8), c = (i % (HID / 8)) * 8;
-    *(bf16x8*)&W1s[r][c] = *(const bf16x8*)&W1bf[r * HID + c];
-  }
-  for (int i = tid; i < HID * (CPAD / 8); i += BLOCK) {
-    const int r = i / (CPAD / 8), c = (i % (CPAD / 8)) * 8;
-    *(bf16x8*)&W2s[r][c] = *(const bf16x8*)&W2bf[r * CPAD + c];
-  }
+  if (tid < HID) L.db1[tid] = 0.f;
+  if (tid < CPAD) L.db2[tid] = 0.f;
+  if (tid == 0) L.loss[0] = 0.f;
+  zero_dl_pad(L);
+  load_weight_images(L, W1bf, W2bf);
+  load_x_chunk(Xbf, L, row0, B);
   __syncthreads();
 
   ChunkAcc acc;
   acc.dW1 = (f32x4){0.f, 0.f, 0.f, 0.f};
   acc.dW2 = (f32x4){0.f, 0.f, 0.f, 0.f};
-  chunk_fwd_bwd(Xs, Hs, DLs, DHs, W1s, W2s, master + OFF_B1, master + OFF_B2,
-                y, row0, B, invBtot, db1_s, db2_s, &loss_s, acc);
+  chunk_fwd_bwd(L, master + OFF_B1, master + OFF_B2, y, row0, B, invBtot, acc);
 
   {
-    const int mt = wave >> 1, nt = wave & 1;
+    const int ht = wave & 1, it = wave >> 1;
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int mrow = mt * 16 + lg * 4 + r;
-      atomicAdd(&grads[OFF_W1 + mrow * HID + nt * 16 + lr], acc.dW1[r]);
+      const int h = ht * 16 + lg * 4 + r;          // D row = h index
+      const int in = it * 16 + lr;                 // D col = input feature
+      atomicAdd(&grads[OFF_W1 + in * HID + h], acc.dW1[r]);
     }
   }
   if (wave < 2) {
@@ -345,9 +437,9 @@ mlp_step_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y, int B,
       atomicAdd(&grads[OFF_W2 + h * CPAD + lr], acc.dW2[r]);
     }
   }
-  if (tid < HID) atomicAdd(&grads[OFF_B1 + tid], db1_s[tid]);
-  else if (tid < HID + CPAD) atomicAdd(&grads[OFF_B2 + tid - HID], db2_s[tid - HID]);
-  else if (tid == HID + CPAD) atomicAdd(&grads[OFF_LOSS], loss_s);
+  if (tid < HID) atomicAdd(&grads[OFF_B1 + tid], L.db1[tid]);
+  else if (tid < HID + CPAD) atomicAdd(&grads[OFF_B2 + tid - HID], L.db2[tid - HID]);
+  else if (tid == HID + CPAD) atomicAdd(&grads[OFF_LOSS], L.loss[0]);
 }
 
 // ---------------------------------------------------------------------------
@@ -362,8 +454,8 @@ adam_step_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
   __shared__ float corr1, corr2;
   if (threadIdx.x == 0) {
     const int t = ++(*t_dev);
-    corr1 = 1.f / (1.f - __powf(beta1, (float)t));
-    corr2 = 1.f / (1.f - __powf(beta2, (float)t));
+    corr1 = fast_rcp(1.f - __powf(beta1, (float)t));
+    corr2 = fast_rcp(1.f - __powf(beta2, (float)t));
   }
   __syncthreads();
   for (int i = threadIdx.x; i < NPARAM; i += 256) {
@@ -372,37 +464,15 @@ adam_step_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
     const float vi = beta2 * v[i] + (1.f - beta2) * g * g;
     m[i] = mi;
     v[i] = vi;
-    const float p = master[i] - lr * (mi * corr1) / (sqrtf(vi * corr2) + eps);
+    const float p = master[i] - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
     master[i] = p;
     bfmirror[i] = f2bf(p);
   }
 }
 
 // ---------------------------------------------------------------------------
-// THE single-GPU flagship: persistent multi-step kernel.
-// ONE workgroup (8 waves). Dynamic LDS layout (16 B-aligned carves):
-//   master_s [NPARAM] f32      | m_s [NPARAM] f32 | v_s [NPARAM] f32
-//   W1s [IN][HS] bf16 | W2s [HID][W2S] bf16
-//   Xs [ROWS][XS] | Hs/DLs/DHs [ROWS][HS] bf16
-//   db1_s[HID] db2_s[CPAD] loss_s[4] f32
-// Total ~87 KB — needs hipFuncAttributeMaxDynamicSharedMemorySize (set by
-// the launcher); single workgroup, so 1-block-per-CU residency is free.
+// persistent multi-step kernel: ONE workgroup, optimizer state in LDS
 // ---------------------------------------------------------------------------
-
-#define ALIGN16(x) (((x) + 15) & ~15)
-#define LDS_MASTER 0
-#define LDS_M      ALIGN16(LDS_MASTER + NPARAM * 4)
-#define LDS_V      ALIGN16(LDS_M + NPARAM * 4)
-#define LDS_W1S    ALIGN16(LDS_V + NPARAM * 4)
-#define LDS_W2S    ALIGN16(LDS_W1S + IN * HS * 2)
-#define LDS_XS     ALIGN16(LDS_W2S + HID * W2S * 2)
-#define LDS_HS     ALIGN16(LDS_XS + ROWS * XS * 2)
-#define LDS_DLS    ALIGN16(LDS_HS + ROWS * HS * 2)
-#define LDS_DHS    ALIGN16(LDS_DLS + ROWS * HS * 2)
-#define LDS_DB1    ALIGN16(LDS_DHS + ROWS * HS * 2)
-#define LDS_DB2    ALIGN16(LDS_DB1 + HID * 4)
-#define LDS_LOSS   ALIGN16(LDS_DB2 + CPAD * 4)
-#define LDS_TOTAL  ALIGN16(LDS_LOSS + 16)
 
 extern "C" __global__ void __launch_bounds__(BLOCK, 1)
 mlp_train_steps_kernel(const u16* __restrict__ Xbf,  // [N][IN] staged bf16
@@ -416,91 +486,87 @@ mlp_train_steps_kernel(const u16* __restrict__ Xbf,  // [N][IN] staged bf16
                        float* __restrict__ loss_out, // last-step loss
                        float lr, float beta1, float beta2, float eps) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* master_s = (float*)(smem + LDS_MASTER);
-  float* m_s = (float*)(smem + LDS_M);
-  float* v_s = (float*)(smem + LDS_V);
-  u16 (*W1s)[HS] = (u16(*)[HS])(smem + LDS_W1S);
-  u16 (*W2s)[W2S] = (u16(*)[W2S])(smem + LDS_W2S);
-  u16 (*Xs)[XS] = (u16(*)[XS])(smem + LDS_XS);
-  u16 (*Hs)[HS] = (u16(*)[HS])(smem + LDS_HS);
-  u16 (*DLs)[HS] = (u16(*)[HS])(smem + LDS_DLS);
-  u16 (*DHs)[HS] = (u16(*)[HS])(smem + LDS_DHS);
-  float* db1_s = (float*)(smem + LDS_DB1);
-  float* db2_s = (float*)(smem + LDS_DB2);
-  float* loss_s = (float*)(smem + LDS_LOSS);
+  const Lds L = carve(smem);
+  float* master_s = (float*)(smem + C_MASTER);
+  float* m_s = (float*)(smem + C_M);
+  float* v_s = (float*)(smem + C_V);
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int l = tid & 63;
-  const int lg = l >> 4, ln = l & 15;   // ln: lane column (lr collides with the lr arg)
+  const int lg = l >> 4, ln = l & 15;
 
-  // ---- load optimizer state + weights into LDS ------------------------------
+  // ---- optimizer state + weight images into LDS -----------------------------
   for (int i = tid; i < NPARAM; i += BLOCK) {
-    const float p = master[i];
-    master_s[i] = p;
+    master_s[i] = master[i];
     m_s[i] = m[i];
     v_s[i] = v[i];
-    // fill bf16 weight images from the master (single source of truth)
-    if (i < OFF_B1) W1s[i / HID][i % HID] = f2bf(p);
-    else if (i >= OFF_W2 && i < OFF_B2) {
-      const int j = i - OFF_W2;
-      W2s[j / CPAD][j % CPAD] = f2bf(p);
-    }
   }
-  zero_dl_pad(DLs);
+  __syncthreads();
+  // weight images from the fp32 master (single source of truth)
+  for (int i = tid; i < HID * IN; i += BLOCK) {
+    const int h = i / IN, k = i % IN;
+    L.W1T[h][k] = f2bf(master_s[OFF_W1 + k * HID + h]);
+  }
+  for (int i = tid; i < HID * 32; i += BLOCK) {
+    const int h = i / 32, c = i % 32;
+    L.W2s[h][c] = (c < CPAD) ? f2bf(master_s[OFF_W2 + h * CPAD + c]) : (u16)0;
+  }
+  for (int i = tid; i < CPAD * HID; i += BLOCK) {
+    const int c = i / HID, h = i % HID;
+    L.W2T[c][h] = f2bf(master_s[OFF_W2 + h * CPAD + c]);
+  }
+  zero_dl_pad(L);
   const int t0 = *t_dev;
-  // per-thread incremental bias-correction powers
   float b1t = __powf(beta1, (float)t0), b2t = __powf(beta2, (float)t0);
   __syncthreads();
 
   const int batches = (int)(N / B);
   const int chunks = B / ROWS;
+  const float invB = fast_rcp((float)B);
 
   for (int s = 0; s < n_steps; ++s) {
     const long long base = (long long)(s % batches) * B;
 
-    if (tid < HID) db1_s[tid] = 0.f;
-    if (tid < CPAD) db2_s[tid] = 0.f;
-    if (tid == 0) loss_s[0] = 0.f;
+    if (tid < HID) L.db1[tid] = 0.f;
+    if (tid < CPAD) L.db2[tid] = 0.f;
+    if (tid == 0) L.loss[0] = 0.f;
 
     ChunkAcc acc;
     acc.dW1 = (f32x4){0.f, 0.f, 0.f, 0.f};
     acc.dW2 = (f32x4){0.f, 0.f, 0.f, 0.f};
-    const float invB = 1.f / (float)B;
 
     for (int ch = 0; ch < chunks; ++ch) {
-      const int row0 = (int)base + ch * ROWS;
-      load_x_chunk(Xbf + 0, Xs, row0, (int)N);   // rows always < N here
+      const long long row0 = base + (long long)ch * ROWS;
+      load_x_chunk(Xbf, L, row0, N);
       __syncthreads();
-      chunk_fwd_bwd(Xs, Hs, DLs, DHs, W1s, W2s, master_s + OFF_B1,
-                    master_s + OFF_B2, y, row0, (int)N, invB,
-                    db1_s, db2_s, loss_s, acc);
+      chunk_fwd_bwd(L, master_s + OFF_B1, master_s + OFF_B2, y, row0, N, invB, acc);
     }
 
     // ---- fused in-LDS Adam --------------------------------------------------
     b1t *= beta1;
     b2t *= beta2;
-    const float corr1 = 1.f / (1.f - b1t);
-    const float corr2 = 1.f / (1.f - b2t);
+    const float corr1 = fast_rcp(1.f - b1t);
+    const float corr2 = fast_rcp(1.f - b2t);
 
-    // helper lambda-ish macro: update index i with grad g
     #define ADAM_UPD(i, g)                                                    \
       {                                                                       \
         const float mi = beta1 * m_s[i] + (1.f - beta1) * (g);                \
         const float vi = beta2 * v_s[i] + (1.f - beta2) * (g) * (g);          \
         m_s[i] = mi;                                                          \
         v_s[i] = vi;                                                          \
-        master_s[i] = master_s[i] - lr * (mi * corr1) / (sqrtf(vi * corr2) + eps); \
+        master_s[i] -= lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps); \
       }
 
     {
-      const int mt = wave >> 1, nt = wave & 1;
+      const int ht = wave & 1, it = wave >> 1;
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int mrow = mt * 16 + lg * 4 + r;
-        const int idx = OFF_W1 + mrow * HID + nt * 16 + ln;
+        const int h = ht * 16 + lg * 4 + r;
+        const int in = it * 16 + ln;
+        const int idx = OFF_W1 + in * HID + h;
         ADAM_UPD(idx, acc.dW1[r]);
-        W1s[mrow][nt * 16 + ln] = f2bf(master_s[idx]);
+        L.W1T[h][in] = f2bf(master_s[idx]);
       }
     }
     if (wave < 2) {
@@ -509,15 +575,16 @@ mlp_train_steps_kernel(const u16* __restrict__ Xbf,  // [N][IN] staged bf16
         const int h = wave * 16 + lg * 4 + r;
         const int idx = OFF_W2 + h * CPAD + ln;
         ADAM_UPD(idx, acc.dW2[r]);
-        W2s[h][ln] = f2bf(master_s[idx]);
+        const u16 wb = f2bf(master_s[idx]);
+        L.W2s[h][ln] = wb;
+        L.W2T[ln][h] = wb;
       }
     }
     if (wave == 2) {
-      // biases: lanes 0..31 -> b1, lanes 32..47 -> b2
       if (l < HID) {
-        ADAM_UPD(OFF_B1 + l, db1_s[l]);
+        ADAM_UPD(OFF_B1 + l, L.db1[l]);
       } else if (l < HID + CPAD) {
-        ADAM_UPD(OFF_B2 + (l - HID), db2_s[l - HID]);
+        ADAM_UPD(OFF_B2 + (l - HID), L.db2[l - HID]);
       }
     }
     #undef ADAM_UPD
@@ -527,7 +594,7 @@ mlp_train_steps_kernel(const u16* __restrict__ Xbf,  // [N][IN] staged bf16
   // ---- write state back to HBM ----------------------------------------------
   if (tid == 0) {
     *t_dev = t0 + n_steps;
-    *loss_out = loss_s[0];
+    *loss_out = L.loss[0];
   }
   for (int i = tid; i < NPARAM; i += BLOCK) {
     master[i] = master_s[i];
@@ -538,7 +605,10 @@ mlp_train_steps_kernel(const u16* __restrict__ Xbf,  // [N][IN] staged bf16
 }
 
 // ---------------------------------------------------------------------------
-// fused predict: standardize + fwd + argmax (serving hot path)
+// fused predict: standardize + fwd + argmax (serving hot path).
+// Uses the same transposed-output fragment scheme (logits arrive as L^T,
+// so the class axis lives in registers/lane-groups and the row-argmax is a
+// 2-level shuffle).
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void __launch_bounds__(BLOCK)
@@ -549,86 +619,89 @@ mlp_predict_kernel(const float* __restrict__ X, int B,
                    const float* __restrict__ master,
                    int* __restrict__ preds,
                    float* __restrict__ probs /* optional [B][CLS] */) {
-  __shared__ u16 Xs[ROWS][XS];
-  __shared__ u16 Hs[ROWS][HS];
-  __shared__ u16 W1s[IN][HS];
-  __shared__ u16 W2s[HID][W2S];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const Lds L = carve(smem);
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int l = tid & 63;
   const int lg = l >> 4, lr = l & 15;
-  const int row0 = blockIdx.x * ROWS;
+  const long long row0 = (long long)blockIdx.x * ROWS;
   const int wrow = wave * 16;
 
+  // standardize on load (fp32 -> bf16); only Xs needed (no bwd)
   for (int i = tid; i < ROWS * IN; i += BLOCK) {
     const int r = i / IN, c = i % IN;
     const float v =
-        (row0 + r < B) ? (X[(long long)(row0 + r) * IN + c] - mean[c]) * invstd[c] : 0.f;
-    Xs[r][c] = f2bf(v);
+        (row0 + r < B) ? (X[(row0 + r) * IN + c] - mean[c]) * invstd[c] : 0.f;
+    L.Xs[r][c] = f2bf(v);
   }
-  for (int i = tid; i < IN * (HID / 8); i += BLOCK) {
-    const int r = i / (HID / 8), c = (i % (HID / 8)) * 8;
-    *(bf16x8*)&W1s[r][c] = *(const bf16x8*)&W1bf[r * HID + c];
-  }
-  for (int i = tid; i < HID * (CPAD / 8); i += BLOCK) {
-    const int r = i / (CPAD / 8), c = (i % (CPAD / 8)) * 8;
-    *(bf16x8*)&W2s[r][c] = *(const bf16x8*)&W2bf[r * CPAD + c];
-  }
+  load_weight_images(L, W1bf, W2bf);
   __syncthreads();
 
   const float* b1 = master + OFF_B1;
   const float* b2 = master + OFF_B2;
 
-  for (int nt = 0; nt < HID / 16; ++nt) {
+  // fwd1: H^T = W1T @ B(Xs) -> write Hs only
+  #pragma unroll
+  for (int mt = 0; mt < HID / 16; ++mt) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
     for (int ks = 0; ks < IN / 32; ++ks) {
-      const bf16x8 a = *(const bf16x8*)&Xs[wrow + lr][ks * 32 + lg * 8];
-      bf16x8 b;
-      #pragma unroll
-      for (int i = 0; i < 8; ++i) b[i] = (short)W1s[ks * 32 + lg * 8 + i][nt * 16 + lr];
+      const bf16x8 a = *(const bf16x8*)&L.W1T[mt * 16 + lr][ks * 32 + lg * 8];
+      const bf16x8 b = *(const bf16x8*)&L.Xs[wrow + lr][ks * 32 + lg * 8];
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float h = acc[r] + b1[nt * 16 + lr];
-      Hs[wrow + lg * 4 + r][nt * 16 + lr] = f2bf(h > 0.f ? h : 0.f);
+      const int h = mt * 16 + lg * 4 + r;
+      float hv = acc[r] + b1[h];
+      L.Hs[wrow + lr][h] = f2bf(hv > 0.f ? hv : 0.f);
     }
   }
   __syncthreads();
 
+  // fwd2: L^T = W2T @ B(Hs); argmax/softmax over the class axis
   {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    const bf16x8 a = *(const bf16x8*)&Hs[wrow + lr][lg * 8];
-    bf16x8 b;
-    #pragma unroll
-    for (int i = 0; i < 8; ++i) b[i] = (short)W2s[lg * 8 + i][lr];
+    const bf16x8 a = *(const bf16x8*)&L.W2T[lr][lg * 8];
+    const bf16x8 b = *(const bf16x8*)&L.Hs[wrow + lr][lg * 8];
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
 
+    const int row = wrow + lr;
+    float best = -1e30f;
+    int bcol = CLS;
+    float logit[4];
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int row = wrow + lg * 4 + r;
-      float logit = acc[r] + b2[lr];
-      if (lr >= CLS) logit = -1e30f;
-      // argmax across the row's 16 lanes; ties pick the lowest column
-      float best = logit;
-      int bcol = lr;
+      const int c = lg * 4 + r;
+      logit[r] = (c < CLS) ? acc[r] + b2[c] : -1e30f;
+      // ties pick the lowest class (match torch.argmax)
+      if (logit[r] > best) { best = logit[r]; bcol = c; }
+    }
+    #pragma unroll
+    for (int d = 16; d < 64; d <<= 1) {
+      const float ov = __shfl_xor(best, d, 64);
+      const int oc = __shfl_xor(bcol, d, 64);
+      if (ov > best || (ov == best && oc < bcol)) { best = ov; bcol = oc; }
+    }
+    if (lg == 0 && row0 + row < B) preds[row0 + row] = bcol;
+    if (probs != nullptr) {
+      float s = 0.f, e[4];
       #pragma unroll
-      for (int d = 1; d < 16; d <<= 1) {
-        const float ov = __shfl_xor(best, d, 64);
-        const int oc = __shfl_xor(bcol, d, 64);
-        if (ov > best || (ov == best && oc < bcol)) { best = ov; bcol = oc; }
+      for (int r = 0; r < 4; ++r) {
+        const int c = lg * 4 + r;
+        e[r] = (c < CLS) ? __expf(logit[r] - best) : 0.f;
+        s += e[r];
       }
-      if (lr == 0 && row0 + row < B) preds[row0 + row] = bcol;
-      if (probs != nullptr) {
-        // shuffle reductions run with ALL lanes active; only the write is
-        // guarded (an inactive lane's shfl result is undefined)
-        const float e = __expf(logit - best);   // best == row max
-        float s = e;
-        #pragma unroll
-        for (int d = 1; d < 16; d <<= 1) s += __shfl_xor(s, d, 64);
-        if (lr < CLS && row0 + row < B) {
-          probs[(long long)(row0 + row) * CLS + lr] = e / s;
+      s += __shfl_xor(s, 16, 64);
+      s += __shfl_xor(s, 32, 64);
+      const float rs = fast_rcp(s);
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int c = lg * 4 + r;
+        if (c < CLS && row0 + row < B) {
+          probs[(row0 + row) * CLS + c] = e[r] * rs;
         }
       }
     }
@@ -638,6 +711,11 @@ mlp_predict_kernel(const float* __restrict__ X, int B,
 // ---------------------------------------------------------------------------
 // host launchers (extern "C"; stream-ordered, capture-safe)
 // ---------------------------------------------------------------------------
+
+static int set_lds(const void* kernel, int bytes) {
+  return (int)hipFuncSetAttribute(kernel, hipFuncAttributeMaxDynamicSharedMemorySize,
+                                  bytes);
+}
 
 extern "C" {
 
@@ -661,9 +739,11 @@ void launch_mlp_step(const unsigned short* Xbf, const int* y, int B,
                      const unsigned short* W1bf, const unsigned short* W2bf,
                      const float* master, float* grads, float invBtot,
                      hipStream_t stream) {
+  static int done = 0;
+  if (!done) { set_lds((const void*)mlp_step_kernel, C_IMG_TOTAL); done = 1; }
   const int blocks = (B + ROWS - 1) / ROWS;
-  hipLaunchKernelGGL(mlp_step_kernel, dim3(blocks), dim3(BLOCK), 0, stream,
-                     Xbf, y, B, W1bf, W2bf, master, grads, invBtot);
+  hipLaunchKernelGGL(mlp_step_kernel, dim3(blocks), dim3(BLOCK), C_IMG_TOTAL,
+                     stream, Xbf, y, B, W1bf, W2bf, master, grads, invBtot);
 }
 
 int launch_mlp_train_steps(const unsigned short* Xbf, const int* y, long long N,
@@ -672,15 +752,12 @@ int launch_mlp_train_steps(const unsigned short* Xbf, const int* y, long long N,
                            int* t_dev, float* loss_out, float lr, float beta1,
                            float beta2, float eps, hipStream_t stream) {
   if (B % ROWS != 0 || N % B != 0) return -1;   // caller falls back
-  static int lds_ok = 0;
-  if (!lds_ok) {
-    hipError_t err = hipFuncSetAttribute(
-        (const void*)mlp_train_steps_kernel,
-        hipFuncAttributeMaxDynamicSharedMemorySize, LDS_TOTAL);
-    if (err != hipSuccess) return -2;
-    lds_ok = 1;
+  static int done = 0;
+  if (!done) {
+    if (set_lds((const void*)mlp_train_steps_kernel, C_STEPS_TOTAL) != 0) return -2;
+    done = 1;
   }
-  hipLaunchKernelGGL(mlp_train_steps_kernel, dim3(1), dim3(BLOCK), LDS_TOTAL,
+  hipLaunchKernelGGL(mlp_train_steps_kernel, dim3(1), dim3(BLOCK), C_STEPS_TOTAL,
                      stream, Xbf, y, N, B, n_steps, master, bfmirror, m, v,
                      t_dev, loss_out, lr, beta1, beta2, eps);
   return 0;
@@ -690,9 +767,11 @@ void launch_mlp_predict(const float* X, int B, const float* mean,
                         const float* invstd, const unsigned short* W1bf,
                         const unsigned short* W2bf, const float* master,
                         int* preds, float* probs, hipStream_t stream) {
+  static int done = 0;
+  if (!done) { set_lds((const void*)mlp_predict_kernel, C_IMG_TOTAL); done = 1; }
   const int blocks = (B + ROWS - 1) / ROWS;
-  hipLaunchKernelGGL(mlp_predict_kernel, dim3(blocks), dim3(BLOCK), 0, stream,
-                     X, B, mean, invstd, W1bf, W2bf, master, preds, probs);
+  hipLaunchKernelGGL(mlp_predict_kernel, dim3(blocks), dim3(BLOCK), C_IMG_TOTAL,
+                     stream, X, B, mean, invstd, W1bf, W2bf, master, preds, probs);
 }
 
 void launch_adam_step(float* master, unsigned short* bfmirror, const float* grads,
