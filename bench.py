@@ -32,11 +32,42 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=2000)
     p.add_argument("--warmup", type=int, default=200)
-    p.add_argument("--batch", type=int, default=512, help="per-GPU batch size")
+    p.add_argument("--batch", type=int, default=2048, help="per-GPU batch size")
     p.add_argument("--minibatches", type=int, default=64, help="distinct minibatches cycled")
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--no-graph", action="store_true")
+    p.add_argument(
+        "--engine",
+        choices=["auto", "fused", "persistent", "stepwise"],
+        default="auto",
+        help="auto: fused single-kernel step (1 GPU) / 3-kernel+RCCL (DP)",
+    )
     return p.parse_args()
+
+
+def _result(args, n_gpus, B, elapsed, engine, loss):
+    return {
+        "metric": "train_samples_per_sec",
+        "value": args.steps * B * n_gpus / elapsed,
+        "unit": "samples/s",
+        "n_gpus": n_gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": elapsed / args.steps * 1000.0,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic",
+        "config": {
+            "model": "digits_mlp_64x32x10",
+            "global_batch": B * n_gpus,
+            "seq_len": None,
+            "parallelism": f"dp{n_gpus}",
+            "engine": engine,
+            "final_loss": loss,
+        },
+    }
 
 
 def main():
@@ -75,19 +106,39 @@ def main():
     Xbf = clf.stage(X)
     invBtot = 1.0 / (B * world)
 
+    engine = args.engine
+    if engine == "auto":
+        engine = "fused" if (use_gpu and world == 1) else "stepwise"
+
     if use_gpu:
         from unionml_amd.ops import hip_ext
+        from unionml_amd.ops.reference import NPARAM as _NP
 
         ext = hip_ext(required=True)
+        loss_out = clf.grads[_NP : _NP + 1]
 
-        def eager_step(off):
-            clf.grads.zero_()
-            ext.mlp_step(Xbf[off : off + B], y[off : off + B], clf.W1bf, clf.W2bf,
-                         clf.master, clf.grads, invBtot)
-            if dist is not None:
-                dist.all_reduce(clf.grads)
-            ext.adam_step(clf.master, clf.bfmirror, clf.grads, clf.m, clf.v,
-                          clf.t_dev, args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS)
+        if engine == "fused":
+            clf._ensure_slabs((B + 127) // 128)
+
+            def eager_step(off):
+                ok = ext.mlp_step_fused(
+                    Xbf[off : off + B], y[off : off + B], clf.W1bf, clf.W2bf,
+                    clf.master, clf.bfmirror, clf.m, clf.v, clf.t_dev,
+                    clf.slabs, clf.counter, loss_out, invBtot,
+                    args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                )
+                assert ok
+
+        else:
+
+            def eager_step(off):
+                clf.grads.zero_()
+                ext.mlp_step(Xbf[off : off + B], y[off : off + B], clf.W1bf,
+                             clf.W2bf, clf.master, clf.grads, invBtot)
+                if dist is not None:
+                    dist.all_reduce(clf.grads)
+                ext.adam_step(clf.master, clf.bfmirror, clf.grads, clf.m, clf.v,
+                              clf.t_dev, args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS)
 
     else:
 
@@ -95,13 +146,8 @@ def main():
             clf._step(Xbf[off : off + B], y[off : off + B], invBtot, args.lr,
                       dist is not None)
 
-    # single-GPU flagship: the persistent multi-step kernel — K optimizer
-    # steps in ONE launch, weights + Adam state resident in LDS.
-    if use_gpu and world == 1 and not args.no_graph:
-        from unionml_amd.ops.reference import NPARAM as _NP
-
-        loss_out = clf.grads[_NP : _NP + 1]
-
+    # persistent single-workgroup engine: K steps in ONE kernel launch
+    if use_gpu and world == 1 and engine == "persistent":
         def run_steps(k):
             ok = ext.mlp_train_steps(
                 Xbf, y, B, k, clf.master, clf.bfmirror, clf.m, clf.v,
@@ -117,32 +163,7 @@ def main():
         elapsed = time.perf_counter() - t0
         loss = float(loss_out.item())
         assert loss == loss and loss < 1e6, f"training diverged: loss={loss}"
-        print(
-            json.dumps(
-                {
-                    "metric": "train_samples_per_sec",
-                    "value": args.steps * B / elapsed,
-                    "unit": "samples/s",
-                    "n_gpus": 1,
-                    "steps": args.steps,
-                    "warmup": args.warmup,
-                    "ms_per_step": elapsed / args.steps * 1000.0,
-                    "higher_is_better": True,
-                    "scaling": "weak",
-                    "vs_baseline": None,
-                    "dtype": "bf16",
-                    "data": "synthetic",
-                    "config": {
-                        "model": "digits_mlp_64x32x10",
-                        "global_batch": B,
-                        "seq_len": None,
-                        "parallelism": "dp1",
-                        "engine": "persistent_steps_kernel",
-                        "final_loss": loss,
-                    },
-                }
-            )
-        )
+        print(json.dumps(_result(args, 1, B, elapsed, "persistent_steps_kernel", loss)))
         return
 
     # warm up communicator + kernels, then capture one hipGraph per minibatch
@@ -195,33 +216,8 @@ def main():
     assert loss == loss and loss < 1e6, f"training diverged: loss={loss}"
 
     if rank == 0:
-        samples_per_sec = args.steps * B * n_gpus / elapsed
-        print(
-            json.dumps(
-                {
-                    "metric": "train_samples_per_sec",
-                    "value": samples_per_sec,
-                    "unit": "samples/s",
-                    "n_gpus": n_gpus,
-                    "steps": args.steps,
-                    "warmup": args.warmup,
-                    "ms_per_step": elapsed / args.steps * 1000.0,
-                    "higher_is_better": True,
-                    "scaling": "weak",
-                    "vs_baseline": None,
-                    "dtype": "bf16",
-                    "data": "synthetic",
-                    "config": {
-                        "model": "digits_mlp_64x32x10",
-                        "global_batch": B * n_gpus,
-                        "seq_len": None,
-                        "parallelism": f"dp{n_gpus}",
-                        "engine": "hipgraph" if graphs is not None else "eager",
-                        "final_loss": loss,
-                    },
-                }
-            )
-        )
+        engine_name = f"{engine}+{'hipgraph' if graphs is not None else 'eager'}"
+        print(json.dumps(_result(args, n_gpus, B, elapsed, engine_name, loss)))
 
     if dist is not None:
         dist.destroy_process_group()

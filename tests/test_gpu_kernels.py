@@ -151,6 +151,43 @@ def test_train_steps_kernel_matches_stepwise(ext, dev):
     assert err_m < 1e-6, f"moment mismatch {err_m}"
 
 
+def test_step_fused_matches_stepwise(ext, dev):
+    """The fully-fused step (slab reduction + Adam in-kernel via the G16
+    ticket hand-off) must reproduce the 3-kernel sequence."""
+    from unionml_amd.ops import reference as ref
+    from unionml_amd.ops.tabular import ADAM_BETA1, ADAM_BETA2, ADAM_EPS, TabularMLP
+
+    torch.manual_seed(21)
+    B, n_steps = 640, 7   # 5 workgroups, odd batch tail of 0
+    Xbf = (torch.randn(B, 64) * 1.2).bfloat16().to(dev)
+    y = torch.randint(0, 10, (B,), dtype=torch.int32, device=dev)
+
+    a = TabularMLP(device=dev, seed=6)
+    for _ in range(n_steps):
+        a.grads.zero_()
+        ext.mlp_step(Xbf, y, a.W1bf, a.W2bf, a.master, a.grads, 1.0 / B)
+        ext.adam_step(a.master, a.bfmirror, a.grads, a.m, a.v, a.t_dev,
+                      1e-3, ADAM_BETA1, ADAM_BETA2, ADAM_EPS)
+    torch.cuda.synchronize()
+
+    b = TabularMLP(device=dev, seed=6)
+    b._ensure_slabs((B + 127) // 128)
+    loss_out = b.grads[ref.NPARAM : ref.NPARAM + 1]
+    for _ in range(n_steps):
+        ok = ext.mlp_step_fused(Xbf, y, b.W1bf, b.W2bf, b.master, b.bfmirror,
+                                b.m, b.v, b.t_dev, b.slabs, b.counter, loss_out,
+                                1.0 / B, 1e-3, ADAM_BETA1, ADAM_BETA2, ADAM_EPS)
+        assert ok
+    torch.cuda.synchronize()
+
+    assert int(b.t_dev.item()) == n_steps
+    # slab reduction order differs from atomic order -> tiny fp32 drift only
+    err = (a.master - b.master).abs().max().item()
+    assert err < 1e-4, f"master mismatch {err}"
+    a_loss = a.grads[ref.NPARAM].item()
+    assert abs(loss_out.item() - a_loss) < 1e-3
+
+
 def test_train_digits_gpu_accuracy(ext, dev):
     from sklearn.datasets import load_digits
 

@@ -443,6 +443,115 @@ mlp_step_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y, int B,
 }
 
 // ---------------------------------------------------------------------------
+// fully-fused step kernel: fwd + bwd + cross-WG grad reduction + Adam in ONE
+// launch. Each workgroup writes its complete partial-grad slab (plain
+// stores), publishes it with the agent-scope release + ticket-counter
+// hand-off of guide §6 G16 (split-K seam form), and the LAST-arriving
+// workgroup acquires, reduces the slabs and applies Adam. Single-GPU path
+// only (the DP path needs the summed grads for RCCL and keeps
+// mlp_step_kernel + adam_step_kernel).
+// ---------------------------------------------------------------------------
+
+#define SLAB 2624   // NPARAM + loss, padded to a 16-float multiple
+
+extern "C" __global__ void __launch_bounds__(BLOCK)
+mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
+                      int B,
+                      const u16* __restrict__ W1bf, const u16* __restrict__ W2bf,
+                      float* __restrict__ master, u16* __restrict__ bfmirror,
+                      float* __restrict__ m, float* __restrict__ v,
+                      int* __restrict__ t_dev,
+                      float* __restrict__ slabs,    // [n_wg][SLAB]
+                      unsigned* __restrict__ counter,  // zeroed at alloc; self-resetting
+                      float* __restrict__ loss_out,
+                      float invBtot, float lr, float beta1, float beta2,
+                      float eps) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const Lds L = carve(smem);
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int l = tid & 63;
+  const int lg = l >> 4, lr_ = l & 15;
+  const long long row0 = (long long)blockIdx.x * ROWS;
+  const int n_wg = gridDim.x;
+
+  if (tid < HID) L.db1[tid] = 0.f;
+  if (tid < CPAD) L.db2[tid] = 0.f;
+  if (tid == 0) L.loss[0] = 0.f;
+  zero_dl_pad(L);
+  load_weight_images(L, W1bf, W2bf);
+  load_x_chunk(Xbf, L, row0, B);
+  __syncthreads();
+
+  ChunkAcc acc;
+  acc.dW1 = (f32x4){0.f, 0.f, 0.f, 0.f};
+  acc.dW2 = (f32x4){0.f, 0.f, 0.f, 0.f};
+  chunk_fwd_bwd(L, master + OFF_B1, master + OFF_B2, y, row0, B, invBtot, acc);
+
+  // ---- write this WG's complete partial slab (plain stores, no atomics) ----
+  float* slab = slabs + (long long)blockIdx.x * SLAB;
+  {
+    const int ht = wave & 1, it = wave >> 1;
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int h = ht * 16 + lg * 4 + r;
+      const int in = it * 16 + lr_;
+      slab[OFF_W1 + in * HID + h] = acc.dW1[r];
+    }
+  }
+  if (wave < 2) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int h = wave * 16 + lg * 4 + r;
+      slab[OFF_W2 + h * CPAD + lr_] = acc.dW2[r];
+    }
+  }
+  if (tid < HID) slab[OFF_B1 + tid] = L.db1[tid];
+  else if (tid < HID + CPAD) slab[OFF_B2 + tid - HID] = L.db2[tid - HID];
+  else if (tid == HID + CPAD) slab[OFF_LOSS] = L.loss[0];
+
+  // ---- publish (guide §6 G16 R1, plain-store + counter form) ---------------
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");   // EVERY storing wave drains
+  __syncthreads();
+  if (tid == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // keep: ROCm may drop it
+    const unsigned ticket = __hip_atomic_fetch_add(
+        counter, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    L.loss[1] = (ticket == (unsigned)(n_wg - 1)) ? 1.f : 0.f;  // existing LDS obj
+  }
+  __syncthreads();
+  if (L.loss[1] == 0.f) return;   // not the last arriver
+
+  // ---- last WG: acquire, reduce slabs, fused Adam ---------------------------
+  if (tid == 0) {
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    const int t = ++(*t_dev);
+    L.loss[2] = fast_rcp(1.f - __powf(beta1, (float)t));
+    L.loss[3] = fast_rcp(1.f - __powf(beta2, (float)t));
+    *counter = 0u;                 // reset for the next launch (stream-ordered)
+  }
+  __syncthreads();
+  const float corr1 = L.loss[2], corr2 = L.loss[3];
+  for (int i = tid; i <= NPARAM; i += BLOCK) {
+    float g = 0.f;
+    for (int w = 0; w < n_wg; ++w) g += slabs[(long long)w * SLAB + i];
+    if (i == NPARAM) {
+      *loss_out = g;
+      continue;
+    }
+    const float mi = beta1 * m[i] + (1.f - beta1) * g;
+    const float vi = beta2 * v[i] + (1.f - beta2) * g * g;
+    m[i] = mi;
+    v[i] = vi;
+    const float p = master[i] - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
+    master[i] = p;
+    bfmirror[i] = f2bf(p);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // fused Adam (per-step DP path) — single block, device step counter
 // ---------------------------------------------------------------------------
 
@@ -744,6 +853,26 @@ void launch_mlp_step(const unsigned short* Xbf, const int* y, int B,
   const int blocks = (B + ROWS - 1) / ROWS;
   hipLaunchKernelGGL(mlp_step_kernel, dim3(blocks), dim3(BLOCK), C_IMG_TOTAL,
                      stream, Xbf, y, B, W1bf, W2bf, master, grads, invBtot);
+}
+
+int launch_mlp_step_fused(const unsigned short* Xbf, const int* y, int B,
+                          const unsigned short* W1bf, const unsigned short* W2bf,
+                          float* master, unsigned short* bfmirror, float* m,
+                          float* v, int* t_dev, float* slabs, unsigned* counter,
+                          float* loss_out, float invBtot, float lr, float beta1,
+                          float beta2, float eps, int max_slabs,
+                          hipStream_t stream) {
+  const int blocks = (B + ROWS - 1) / ROWS;
+  if (blocks > max_slabs) return -1;
+  static int done = 0;
+  if (!done) {
+    if (set_lds((const void*)mlp_step_fused_kernel, C_IMG_TOTAL) != 0) return -2;
+    done = 1;
+  }
+  hipLaunchKernelGGL(mlp_step_fused_kernel, dim3(blocks), dim3(BLOCK), C_IMG_TOTAL,
+                     stream, Xbf, y, B, W1bf, W2bf, master, bfmirror, m, v, t_dev,
+                     slabs, counter, loss_out, invBtot, lr, beta1, beta2, eps);
+  return 0;
 }
 
 int launch_mlp_train_steps(const unsigned short* Xbf, const int* y, long long N,

@@ -16,6 +16,11 @@ void launch_mlp_step(const unsigned short*, const int*, int, const unsigned shor
 int launch_mlp_train_steps(const unsigned short*, const int*, long long, int, int,
                            float*, unsigned short*, float*, float*, int*, float*,
                            float, float, float, float, hipStream_t);
+int launch_mlp_step_fused(const unsigned short*, const int*, int,
+                          const unsigned short*, const unsigned short*, float*,
+                          unsigned short*, float*, float*, int*, float*, unsigned*,
+                          float*, float, float, float, float, float, int,
+                          hipStream_t);
 void launch_mlp_predict(const float*, int, const float*, const float*,
                         const unsigned short*, const unsigned short*, const float*,
                         int*, float*, hipStream_t);
@@ -83,6 +88,32 @@ void mlp_step(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
                   grads.data_ptr<float>(), (float)invBtot, current_stream());
 }
 
+bool mlp_step_fused(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
+                    torch::Tensor W2bf, torch::Tensor master, torch::Tensor bfmirror,
+                    torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
+                    torch::Tensor slabs, torch::Tensor counter,
+                    torch::Tensor loss_out, double invBtot, double lr,
+                    double beta1, double beta2, double eps) {
+  check(Xbf, torch::kBFloat16, "Xbf");
+  check(y, torch::kInt32, "y");
+  check(master, torch::kFloat32, "master");
+  check(bfmirror, torch::kBFloat16, "bfmirror");
+  check(slabs, torch::kFloat32, "slabs");
+  check(counter, torch::kUInt32, "counter");
+  check(loss_out, torch::kFloat32, "loss_out");
+  TORCH_CHECK(Xbf.size(1) == 64, "IN must be 64");
+  TORCH_CHECK(slabs.dim() == 2 && slabs.size(1) == 2624, "slabs must be [n][2624]");
+  const int rc = launch_mlp_step_fused(
+      bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0), bf16_ptr(W1bf),
+      bf16_ptr(W2bf), master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
+      m.data_ptr<float>(), v.data_ptr<float>(), t_dev.data_ptr<int>(),
+      slabs.data_ptr<float>(), (unsigned*)counter.data_ptr(),
+      loss_out.data_ptr<float>(), (float)invBtot, (float)lr, (float)beta1,
+      (float)beta2, (float)eps, (int)slabs.size(0), current_stream());
+  TORCH_CHECK(rc != -2, "mlp_step_fused: hipFuncSetAttribute(LDS) failed");
+  return rc == 0;
+}
+
 bool mlp_train_steps(torch::Tensor Xbf, torch::Tensor y, int64_t batch,
                      int64_t n_steps, torch::Tensor master, torch::Tensor bfmirror,
                      torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
@@ -142,6 +173,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("standardize_fit", &standardize_fit, "column mean/invstd (CDNA4)");
   m.def("standardize_apply", &standardize_apply, "(x-mean)*invstd -> bf16 (CDNA4)");
   m.def("mlp_step", &mlp_step, "fused MLP fwd+bwd step (CDNA4 MFMA)");
+  m.def("mlp_step_fused", &mlp_step_fused,
+        "fully-fused step: fwd+bwd + cross-WG slab reduction + Adam, one launch");
   m.def("mlp_train_steps", &mlp_train_steps,
         "persistent multi-step training kernel (weights+Adam resident in LDS)");
   m.def("mlp_predict", &mlp_predict, "fused standardize+fwd+argmax (CDNA4 MFMA)");

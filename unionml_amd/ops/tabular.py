@@ -52,8 +52,15 @@ class TabularMLP:
         self.grads = torch.zeros(NPARAM + 1, dtype=torch.float32, device=self.device)
         self.mean = torch.zeros(IN, dtype=torch.float32, device=self.device)
         self.invstd = torch.ones(IN, dtype=torch.float32, device=self.device)
+        self.slabs = None       # per-WG partial-grad slabs for the fused step
+        self.counter = None     # G16 ticket counter (self-resetting)
         self._graph = None
         self._graph_key = None
+
+    def _ensure_slabs(self, n_wg: int):
+        if self.slabs is None or self.slabs.shape[0] < n_wg:
+            self.slabs = torch.zeros(n_wg, 2624, dtype=torch.float32, device=self.device)
+            self.counter = torch.zeros(1, dtype=torch.uint32, device=self.device)
 
     # -- views ----------------------------------------------------------------
 
@@ -111,6 +118,42 @@ class TabularMLP:
             )
         return self.grads[NPARAM]
 
+    def _train_epochs_fused(self, Xbf, y, batches, *, epochs, lr, use_graph) -> float:
+        ext = hip_ext()
+        self._ensure_slabs(max((bs + 127) // 128 for _, bs in batches))
+        loss_out = self.grads[NPARAM : NPARAM + 1]
+
+        def run_epoch():
+            for off, bs in batches:
+                ok = ext.mlp_step_fused(
+                    Xbf[off : off + bs], y[off : off + bs], self.W1bf, self.W2bf,
+                    self.master, self.bfmirror, self.m, self.v, self.t_dev,
+                    self.slabs, self.counter, loss_out, 1.0 / bs,
+                    lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                )
+                assert ok, "mlp_step_fused slab capacity exceeded"
+
+        run_epoch()   # warmup epoch (eager)
+        remaining = epochs - 1
+        key = ("fused", len(batches), lr, Xbf.data_ptr(), y.data_ptr())
+        if use_graph and self._graph_key != key:
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    run_epoch()
+                self._graph, self._graph_key = g, key
+            except RuntimeError as exc:
+                logger.warning("hipGraph capture failed (%s); eager stepping", exc)
+                self._graph, self._graph_key = None, None
+        if use_graph and self._graph is not None:
+            for _ in range(remaining):
+                self._graph.replay()
+        else:
+            for _ in range(remaining):
+                run_epoch()
+        torch.cuda.synchronize(self.device)
+        return float(loss_out.item())
+
     def train_epochs(
         self,
         Xbf: torch.Tensor,
@@ -121,6 +164,7 @@ class TabularMLP:
         lr: float = 1e-3,
         use_graph: bool = True,
         world_size: int = 1,
+        engine: str = "auto",
     ) -> float:
         """Minibatch Adam training over the staged (bf16) features.
 
@@ -134,25 +178,24 @@ class TabularMLP:
         ]
         allreduce = world_size > 1
 
-        # single-GPU flagship: the persistent multi-step kernel (weights +
-        # Adam state resident in LDS, zero launches per step). Requires
-        # batch % 128 == 0 and n % batch == 0; otherwise fall through.
-        if (
-            self.use_hip
-            and not allreduce
-            and batch_size % 128 == 0
-            and n % batch_size == 0
-        ):
-            loss_out = self.grads[NPARAM : NPARAM + 1]
-            n_steps = epochs * (n // batch_size)
-            ok = hip_ext().mlp_train_steps(
-                Xbf, y, batch_size, n_steps, self.master, self.bfmirror,
-                self.m, self.v, self.t_dev, loss_out,
-                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+        # single-GPU flagship: the fully-fused step kernel (fwd+bwd +
+        # cross-WG slab reduction + Adam in ONE launch), with the epoch's
+        # minibatch loop captured into a hipGraph.
+        if self.use_hip and not allreduce:
+            if engine == "persistent" and batch_size % 128 == 0 and n % batch_size == 0:
+                loss_out = self.grads[NPARAM : NPARAM + 1]
+                n_steps = epochs * (n // batch_size)
+                ok = hip_ext().mlp_train_steps(
+                    Xbf, y, batch_size, n_steps, self.master, self.bfmirror,
+                    self.m, self.v, self.t_dev, loss_out,
+                    lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                )
+                if ok:
+                    torch.cuda.synchronize(self.device)
+                    return float(loss_out.item())
+            return self._train_epochs_fused(
+                Xbf, y, batches, epochs=epochs, lr=lr, use_graph=use_graph
             )
-            if ok:
-                torch.cuda.synchronize(self.device)
-                return float(loss_out.item())
 
         def run_epoch():
             last = None
