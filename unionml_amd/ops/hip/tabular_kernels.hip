@@ -462,12 +462,14 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
                       float* __restrict__ m, float* __restrict__ v,
                       int* __restrict__ t_dev,
                       float* __restrict__ slabs,    // [n_wg][SLAB]
-                      unsigned* __restrict__ counter,  // zeroed at alloc; self-resetting
+                      unsigned* __restrict__ counter,  // MONOTONIC ticket counter
                       float* __restrict__ loss_out,
                       float invBtot, float lr, float beta1, float beta2,
                       float eps) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const Lds L = carve(smem);
+  // L.loss slots: [0] loss accum, [1] (bits) poll base, [2] t_pre, [3] unused
+  unsigned* lossu = (unsigned*)L.loss;
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -478,7 +480,13 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
 
   if (tid < HID) L.db1[tid] = 0.f;
   if (tid < CPAD) L.db2[tid] = 0.f;
-  if (tid == 0) L.loss[0] = 0.f;
+  if (tid == 0) {
+    L.loss[0] = 0.f;
+    // read the step counter BEFORE any workgroup of this launch can have
+    // advanced it (the writer only writes after every WG has published,
+    // i.e. after every WG has passed this point)
+    L.loss[2] = (float)(*t_dev);
+  }
   zero_dl_pad(L);
   load_weight_images(L, W1bf, W2bf);
   load_x_chunk(Xbf, L, row0, B);
@@ -511,32 +519,61 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   else if (tid < HID + CPAD) slab[OFF_B2 + tid - HID] = L.db2[tid - HID];
   else if (tid == HID + CPAD) slab[OFF_LOSS] = L.loss[0];
 
-  // ---- publish (guide §6 G16 R1, plain-store + counter form) ---------------
+  // ---- publish + all-WG barrier (guide §6 G16: R1 release + per-slab epoch
+  // tag, R2 data-is-the-flag poll). The tag is this step's epoch t_pre+1 —
+  // unique per launch (exactly one WG advances t_dev per launch), so the
+  // scheme is robust to any grid size, graph replay, and engine mixing;
+  // no counter, no reset. ----------------------------------------------------
+  const unsigned epoch = (unsigned)L.loss[2] + 1u;
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");   // EVERY storing wave drains
   __syncthreads();
   if (tid == 0) {
     __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // keep: ROCm may drop it
-    const unsigned ticket = __hip_atomic_fetch_add(
-        counter, 1u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-    L.loss[1] = (ticket == (unsigned)(n_wg - 1)) ? 1.f : 0.f;  // existing LDS obj
-  }
-  __syncthreads();
-  if (L.loss[1] == 0.f) return;   // not the last arriver
-
-  // ---- last WG: acquire, reduce slabs, fused Adam ---------------------------
-  if (tid == 0) {
+    __hip_atomic_store((unsigned*)(slab + NPARAM + 1), epoch, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    // relaxed sweep of every slab's tag word, bounded spin (G16: never
+    // poll with an acquire)
+    unsigned spins = 0;
+    bool ok = true;
+    for (int w = 0; w < n_wg;) {
+      const unsigned tag = __hip_atomic_load(
+          (const unsigned*)(slabs + (long long)w * SLAB + NPARAM + 1),
+          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      if (tag == epoch) {
+        ++w;
+      } else {
+        __builtin_amdgcn_s_sleep(2);
+        if (++spins > 100000000u) { ok = false; break; }
+      }
+    }
     __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    const int t = ++(*t_dev);
-    L.loss[2] = fast_rcp(1.f - __powf(beta1, (float)t));
-    L.loss[3] = fast_rcp(1.f - __powf(beta2, (float)t));
-    *counter = 0u;                 // reset for the next launch (stream-ordered)
+    lossu[1] = ok ? 0u : 1u;
   }
   __syncthreads();
-  const float corr1 = L.loss[2], corr2 = L.loss[3];
-  for (int i = tid; i <= NPARAM; i += BLOCK) {
-    float g = 0.f;
-    for (int w = 0; w < n_wg; ++w) g += slabs[(long long)w * SLAB + i];
+  if (lossu[1] != 0u) {             // timed out: poison the loss, keep going
+    if (tid == 0 && blockIdx.x == 0) *loss_out = __builtin_nanf("");
+    return;
+  }
+
+  // ---- every WG reduces its own param stripe + applies Adam ----------------
+  const float t_new = L.loss[2] + 1.f;
+  const float corr1 = fast_rcp(1.f - __powf(beta1, t_new));
+  const float corr2 = fast_rcp(1.f - __powf(beta2, t_new));
+  const int span = (NPARAM + 1 + n_wg - 1) / n_wg;
+  const int lo = blockIdx.x * span;
+  const int hi = min(lo + span, NPARAM + 1);
+  for (int i = lo + tid; i < hi; i += BLOCK) {
+    float g0 = 0.f, g1 = 0.f, g2 = 0.f, g3 = 0.f;
+    int w = 0;
+    for (; w + 4 <= n_wg; w += 4) {      // 4 independent load chains
+      g0 += slabs[(long long)w * SLAB + i];
+      g1 += slabs[(long long)(w + 1) * SLAB + i];
+      g2 += slabs[(long long)(w + 2) * SLAB + i];
+      g3 += slabs[(long long)(w + 3) * SLAB + i];
+    }
+    for (; w < n_wg; ++w) g0 += slabs[(long long)w * SLAB + i];
+    const float g = (g0 + g1) + (g2 + g3);
     if (i == NPARAM) {
       *loss_out = g;
       continue;
@@ -549,6 +586,7 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
     master[i] = p;
     bfmirror[i] = f2bf(p);
   }
+  if (tid == 0 && blockIdx.x == 0) *t_dev = (int)t_new;
 }
 
 // ---------------------------------------------------------------------------
