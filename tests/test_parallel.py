@@ -1,0 +1,128 @@
+"""Data-parallel tests on CPU: gloo backend, world_size 2.
+
+These cover the distributed path the driver exercises on 8 GPUs at
+round end: the bucketed gradient all-reducer, row sharding, and
+Model.train(dp=2) end-to-end through process spawn.
+"""
+
+import os
+
+import numpy as np
+import pandas as pd
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from unionml_amd.parallel.ddp import GradientAllReducer, shard
+
+
+def test_shard_balance():
+    df = pd.DataFrame({"a": range(10)})
+    parts = [shard(df, r, 3) for r in range(3)]
+    assert [len(p) for p in parts] == [4, 3, 3]
+    assert pd.concat(parts).equals(df)
+    arr = np.arange(7)
+    parts = [shard(arr, r, 2) for r in range(2)]
+    assert sum(len(p) for p in parts) == 7
+
+
+def test_reducer_single_process_noop():
+    from unionml_amd.parallel.ddp import maybe_wrap
+
+    model = torch.nn.Linear(4, 2)
+    assert maybe_wrap(model) is None  # no process group -> no-op
+
+
+def _reducer_worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)  # same init on every rank
+        model = torch.nn.Sequential(
+            torch.nn.Linear(8, 16), torch.nn.ReLU(), torch.nn.Linear(16, 4)
+        )
+        reducer = GradientAllReducer(model, bucket_mb=0.0001)  # force many buckets
+        torch.manual_seed(100 + rank)  # different data per rank
+        x = torch.randn(16, 8)
+        y = torch.randint(0, 4, (16,))
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        reducer.finalize()
+
+        # oracle: average of per-rank grads, computed via explicit gather
+        grads = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+        gathered = [torch.zeros_like(grads) for _ in range(world)]
+        dist.all_gather(gathered, grads)
+        for g in gathered:
+            assert torch.allclose(g, grads, atol=1e-6), "ranks disagree after reduce"
+
+        # compare against a manual recompute of the averaged gradient
+        model2 = torch.nn.Sequential(
+            torch.nn.Linear(8, 16), torch.nn.ReLU(), torch.nn.Linear(16, 4)
+        )
+        torch.manual_seed(0)
+        for p2, p in zip(model2.parameters(), model.parameters()):
+            pass  # weights identical by seed; recompute local grad
+        torch.manual_seed(0)
+        model3 = torch.nn.Sequential(
+            torch.nn.Linear(8, 16), torch.nn.ReLU(), torch.nn.Linear(16, 4)
+        )
+        torch.manual_seed(100 + rank)
+        x3 = torch.randn(16, 8)
+        y3 = torch.randint(0, 4, (16,))
+        loss3 = torch.nn.functional.cross_entropy(model3(x3), y3)
+        loss3.backward()
+        local = torch.cat([p.grad.reshape(-1) for p in model3.parameters()])
+        gathered_local = [torch.zeros_like(local) for _ in range(world)]
+        dist.all_gather(gathered_local, local)
+        expected = torch.stack(gathered_local).mean(dim=0)
+        assert torch.allclose(grads, expected, atol=1e-6), (
+            (grads - expected).abs().max()
+        )
+
+        # zero_grad re-installs views; a second step must also work
+        reducer.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        reducer.finalize()
+        with open(os.path.join(out_dir, f"ok_{rank}"), "w") as f:
+            f.write("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_gradient_allreducer_world2(tmp_path):
+    from unionml_amd.parallel.launch import _free_port
+
+    port = _free_port()
+    mp.start_processes(
+        _reducer_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True,
+        start_method="spawn",
+    )
+    assert (tmp_path / "ok_0").exists() and (tmp_path / "ok_1").exists()
+
+
+def test_model_train_dp2():
+    """Model.train(dp=2) spawns 2 gloo ranks, shards the train split, and
+    returns a rank-0 artifact."""
+    from dp_app import model
+
+    model.artifact = None
+    model_obj, metrics = model.train(dp=2, n=120, trainer_kwargs={"epochs": 2})
+    assert model.artifact is not None
+    assert 0.0 <= metrics["train"] <= 1.0
+    preds = model.predict(features=[{"x1": 0.1, "x2": -0.2, "x3": 0.3}])
+    assert len(preds) == 1
+
+
+def test_tabular_dp2_matches_single():
+    """TabularMLP DP training over gloo must equal single-process training
+    on the concatenated data (same seeds, full-batch steps)."""
+    from dp_app import run_tabular_dp
+
+    single_master, dp_master = run_tabular_dp()
+    assert torch.allclose(single_master, dp_master, rtol=1e-4, atol=1e-6), (
+        (single_master - dp_master).abs().max()
+    )
