@@ -88,6 +88,21 @@ def docker_build_push(image_fqn: str, dockerfile: str = "Dockerfile", context: s
     return image_fqn
 
 
+def _pid_alive(pid: int) -> bool:
+    try:
+        os.kill(pid, 0)
+    except ProcessLookupError:
+        return False
+    except PermissionError:
+        return True
+    # an unreaped child (zombie) is dead for our purposes
+    try:
+        with open(f"/proc/{pid}/stat") as fh:
+            return fh.read().split(") ", 1)[1].split()[0] != "Z"
+    except (FileNotFoundError, IndexError, OSError):
+        return True
+
+
 @dataclass
 class Execution:
     """Handle to a backend execution."""
@@ -299,12 +314,17 @@ class Backend:
         devices = self.gpus.assign(n_gpus)
         if devices is not None:
             env["HIP_VISIBLE_DEVICES"] = devices
-        # make the app module importable in the worker
+        # make the app module AND this framework importable in the worker
+        # (the worker may run from any cwd; pytest tmp dirs, cron, etc.)
+        import unionml_amd
+
+        pythonpath = [str(Path(unionml_amd.__file__).parent.parent)]
         module_file = manifest.get("module_file")
         if module_file:
-            env["PYTHONPATH"] = (
-                str(Path(module_file).parent) + os.pathsep + env.get("PYTHONPATH", "")
-            )
+            pythonpath.insert(0, str(Path(module_file).parent))
+        if env.get("PYTHONPATH"):
+            pythonpath.append(env["PYTHONPATH"])
+        env["PYTHONPATH"] = os.pathsep.join(pythonpath)
 
         with open(exec_dir / "worker.log", "wb") as log:
             proc = subprocess.Popen(
@@ -319,6 +339,14 @@ class Backend:
             proc.wait()
         return execution
 
+    def get_execution(self, execution_id: str) -> Execution:
+        """Rehydrate an execution handle by id (reference fetches from
+        the Flyte admin: remote.py:236-269)."""
+        path = self.root / "executions" / execution_id
+        if not path.is_dir():
+            raise ModelArtifactNotFound(f"execution '{execution_id}' not found")
+        return Execution(id=execution_id, workflow=execution_id.split("-")[0], path=str(path))
+
     def wait(self, execution: Execution, timeout: Optional[float] = None) -> Execution:
         deadline = time.monotonic() + (timeout or 3600)
         while time.monotonic() < deadline:
@@ -329,6 +357,19 @@ class Backend:
                     tail = log.read_text()[-3000:] if log.exists() else ""
                     raise RuntimeError(f"execution {execution.id} FAILED:\n{tail}")
                 return execution
+            # fail fast if the worker died before reaching a terminal status
+            # (e.g. crashed during interpreter startup)
+            pid_file = Path(execution.path) / "pid"
+            if pid_file.exists() and not _pid_alive(int(pid_file.read_text())):
+                # re-check: the worker may have written the status just
+                # before exiting
+                if execution.status not in (STATUS_SUCCEEDED, STATUS_FAILED):
+                    log = Path(execution.path) / "worker.log"
+                    tail = log.read_text()[-3000:] if log.exists() else ""
+                    (Path(execution.path) / "status").write_text(STATUS_FAILED)
+                    raise RuntimeError(
+                        f"execution {execution.id} worker died without a status:\n{tail}"
+                    )
             time.sleep(0.05)
         raise TimeoutError(f"execution {execution.id} did not finish in time")
 
