@@ -1,0 +1,274 @@
+"""FastAPI serving tests.
+
+Route contract mirrors the reference (unionml/fastapi.py:15-70, tested
+in its tests/integration/test_fastapi.py): `GET /` banner, `POST
+/predict` with either features or reader inputs, `GET /health` failing
+without an artifact; plus the subprocess `serve` integration with
+retry-based health polling (reference test_fastapi.py:14-49) and this
+build's dynamic micro-batcher."""
+
+import asyncio
+import json
+import socket
+import subprocess
+import sys
+import time
+from pathlib import Path
+
+import pandas as pd
+import pytest
+from fastapi import FastAPI
+from fastapi.testclient import TestClient
+
+from model_fixtures import build_sklearn_app
+
+
+@pytest.fixture
+def trained_app():
+    model = build_sklearn_app()
+    model.train()
+    app = FastAPI()
+    model.serve(app)
+    return model, app
+
+
+def test_root_banner(trained_app):
+    model, app = trained_app
+    with TestClient(app) as client:
+        resp = client.get("/")
+        assert resp.status_code == 200
+        assert model.name in resp.json()["model"]
+
+
+def test_health_requires_artifact():
+    model = build_sklearn_app()
+    app = FastAPI()
+    model.serve(app)
+    with TestClient(app) as client:
+        assert client.get("/health").status_code == 500
+
+
+def test_health_ok(trained_app):
+    _, app = trained_app
+    with TestClient(app) as client:
+        assert client.get("/health").json() == {"status": "ok"}
+
+
+def test_predict_with_features(trained_app):
+    model, app = trained_app
+    feats = [{"x1": 0.5, "x2": 0.1, "x3": 0.9}, {"x1": 0.2, "x2": 0.8, "x3": 0.4}]
+    with TestClient(app) as client:
+        resp = client.post("/predict", json={"features": feats})
+        assert resp.status_code == 200, resp.text
+        preds = resp.json()
+        assert len(preds) == 2
+        direct = model.predict(features=model._dataset.get_features(feats))
+        assert preds == direct
+
+
+def test_predict_with_reader_inputs(trained_app):
+    _, app = trained_app
+    with TestClient(app) as client:
+        resp = client.post("/predict", json={"inputs": {"n": 7}})
+        assert resp.status_code == 200, resp.text
+        assert len(resp.json()) == 7
+
+
+def test_predict_requires_body(trained_app):
+    _, app = trained_app
+    with TestClient(app) as client:
+        assert client.post("/predict", json={}).status_code == 400
+
+
+def test_predict_no_artifact_500():
+    model = build_sklearn_app()
+    app = FastAPI()
+    model.serve(app)
+    with TestClient(app) as client:
+        resp = client.post("/predict", json={"features": [{"x1": 1, "x2": 2, "x3": 3}]})
+        assert resp.status_code == 500
+
+
+# ----------------------------------------------------------------------
+# dynamic batcher (CPU path of the hipGraph-bucketed serve step)
+# ----------------------------------------------------------------------
+
+
+def test_batcher_coalesces_and_matches_direct():
+    from unionml_amd.serving.batcher import DynamicBatcher
+
+    model = build_sklearn_app()
+    model.train()
+
+    calls = []
+    orig = model._run_predictor
+
+    def counting(model_obj, features):
+        calls.append(len(features))
+        return orig(model_obj, features)
+
+    model._run_predictor = counting
+
+    batcher = DynamicBatcher(model, max_batch_size=64, max_delay_ms=30.0)
+    batcher.start()
+    try:
+        feats = [
+            [{"x1": 0.1 * i, "x2": 0.2, "x3": 0.3}, {"x1": 0.9, "x2": 0.1 * i, "x3": 0.5}]
+            for i in range(8)
+        ]
+
+        async def fire():
+            return await asyncio.gather(*(batcher.submit(f) for f in feats))
+
+        results = asyncio.run(fire())
+    finally:
+        batcher.stop()
+
+    batched_calls = list(calls)
+    assert len(results) == 8
+    for f, r in zip(feats, results):
+        direct = model.predict(features=model._dataset.get_features(f))
+        assert r == direct
+    # 16 rows total, arriving together within max_delay -> far fewer
+    # forwards than requests
+    assert len(batched_calls) < 8, f"batcher did not coalesce: {batched_calls}"
+
+
+def test_batcher_bucketing():
+    from unionml_amd.serving.batcher import bucket_for
+
+    assert [bucket_for(n, 64) for n in (1, 2, 3, 5, 17, 64, 100)] == [1, 2, 4, 8, 32, 64, 64]
+
+
+def test_serving_app_with_batcher_route():
+    model = build_sklearn_app()
+    model.train()
+    app = FastAPI()
+    model.serve(app, batch=True, max_delay_ms=1.0)
+    feats = [{"x1": 0.5, "x2": 0.1, "x3": 0.9}]
+    with TestClient(app) as client:
+        resp = client.post("/predict", json={"features": feats})
+        assert resp.status_code == 200, resp.text
+        assert resp.json() == model.predict(features=model._dataset.get_features(feats))
+
+
+# ----------------------------------------------------------------------
+# subprocess integration: unionml-amd serve + real HTTP
+# ----------------------------------------------------------------------
+
+SERVE_APP = """
+from typing import List
+
+import pandas as pd
+from fastapi import FastAPI
+from sklearn.linear_model import LogisticRegression
+
+from unionml_amd import Dataset, Model
+
+dataset = Dataset(name="serve_ds", targets=["y"], test_size=0.2, shuffle=True, random_state=0)
+model = Model(name="serve_model", init=LogisticRegression, dataset=dataset)
+
+
+@dataset.reader
+def reader(n: int = 50) -> pd.DataFrame:
+    import numpy as np
+
+    rng = np.random.RandomState(3)
+    X = rng.rand(n, 3)
+    return pd.DataFrame(
+        {"x1": X[:, 0], "x2": X[:, 1], "x3": X[:, 2], "y": (X.sum(axis=1) > 1.5).astype(int)}
+    )
+
+
+@model.trainer
+def trainer(
+    estimator: LogisticRegression, features: pd.DataFrame, target: pd.DataFrame
+) -> LogisticRegression:
+    return estimator.fit(features, target.squeeze())
+
+
+@model.predictor
+def predictor(estimator: LogisticRegression, features: pd.DataFrame) -> List[float]:
+    return [float(x) for x in estimator.predict(features)]
+
+
+@model.evaluator
+def evaluator(estimator: LogisticRegression, features: pd.DataFrame, target: pd.DataFrame) -> float:
+    return float((estimator.predict(features) == target.squeeze().to_numpy()).mean())
+
+
+app = FastAPI()
+model.serve(app)
+
+if __name__ == "__main__":
+    model.train()
+    model.save("model.joblib")
+"""
+
+
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+@pytest.mark.timeout(120)
+def test_serve_subprocess_http(tmp_path):
+    import httpx
+
+    (tmp_path / "serve_app.py").write_text(SERVE_APP)
+    # train + save the artifact in-process (reference trains via runpy:
+    # test_fastapi.py:63-78)
+    proc = subprocess.run(
+        [sys.executable, "serve_app.py"], cwd=tmp_path, capture_output=True, text=True
+    )
+    assert proc.returncode == 0, proc.stderr
+    assert (tmp_path / "model.joblib").exists()
+
+    port = _free_port()
+    server = subprocess.Popen(
+        [
+            sys.executable,
+            "-m",
+            "unionml_amd.cli",
+            "serve",
+            "serve_app:app",
+            "--model-path",
+            "model.joblib",
+            "--port",
+            str(port),
+        ],
+        cwd=tmp_path,
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+    )
+    try:
+        # retry-based health polling (reference: test_fastapi.py:30-49)
+        deadline = time.monotonic() + 60
+        healthy = False
+        while time.monotonic() < deadline:
+            if server.poll() is not None:
+                out = server.stdout.read().decode()
+                raise AssertionError(f"server exited early:\n{out[-3000:]}")
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/health", timeout=1.0).status_code == 200:
+                    healthy = True
+                    break
+            except httpx.HTTPError:
+                time.sleep(0.5)
+        assert healthy, "server never became healthy"
+
+        feats = [{"x1": 0.9, "x2": 0.9, "x3": 0.9}, {"x1": 0.0, "x2": 0.1, "x3": 0.0}]
+        resp = httpx.post(
+            f"http://127.0.0.1:{port}/predict", json={"features": feats}, timeout=10.0
+        )
+        assert resp.status_code == 200, resp.text
+        preds = resp.json()
+        assert len(preds) == 2
+        assert all(p in (0.0, 1.0) for p in preds)
+    finally:
+        server.terminate()
+        try:
+            server.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            server.kill()
