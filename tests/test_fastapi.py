@@ -214,13 +214,24 @@ def _free_port() -> int:
 
 @pytest.mark.timeout(120)
 def test_serve_subprocess_http(tmp_path):
+    import os
+
     import httpx
+
+    import unionml_amd
+
+    env = dict(
+        os.environ,
+        PYTHONPATH=str(Path(unionml_amd.__file__).parent.parent)
+        + os.pathsep
+        + os.environ.get("PYTHONPATH", ""),
+    )
 
     (tmp_path / "serve_app.py").write_text(SERVE_APP)
     # train + save the artifact in-process (reference trains via runpy:
     # test_fastapi.py:63-78)
     proc = subprocess.run(
-        [sys.executable, "serve_app.py"], cwd=tmp_path, capture_output=True, text=True
+        [sys.executable, "serve_app.py"], cwd=tmp_path, capture_output=True, text=True, env=env
     )
     assert proc.returncode == 0, proc.stderr
     assert (tmp_path / "model.joblib").exists()
@@ -241,6 +252,7 @@ def test_serve_subprocess_http(tmp_path):
         cwd=tmp_path,
         stdout=subprocess.PIPE,
         stderr=subprocess.STDOUT,
+        env=env,
     )
     try:
         # retry-based health polling (reference: test_fastapi.py:30-49)
