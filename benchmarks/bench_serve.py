@@ -1,0 +1,184 @@
+#!/usr/bin/env python3
+"""Serving benchmark: /predict p50 latency for the digits MLP app
+(BASELINE.md config 3).
+
+Trains the flagship digits MLP (synthetic data, random-init weights),
+serves it through the real FastAPI app (dynamic batcher + bucketed
+hipGraph inference on GPU), and measures:
+
+- sequential single-row POST /predict latency over real HTTP
+  (p50/p90/p99) — the headline "/predict p50";
+- concurrent-load latency with K async clients (batcher coalescing);
+- raw in-process graph-replay latency (no HTTP) for reference.
+
+Prints one JSON line. Run (GPU box):
+    python benchmarks/bench_serve.py --requests 500 --concurrency 16
+"""
+
+import argparse
+import asyncio
+import json
+import os
+import socket
+import statistics
+import sys
+import threading
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--requests", type=int, default=500)
+    p.add_argument("--warmup", type=int, default=50)
+    p.add_argument("--concurrency", type=int, default=16)
+    p.add_argument("--port", type=int, default=0)
+    p.add_argument("--no-batch", action="store_true")
+    return p.parse_args()
+
+
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def pct(xs, q):
+    xs = sorted(xs)
+    return xs[min(len(xs) - 1, int(q * len(xs)))]
+
+
+def main():
+    args = parse_args()
+    import torch
+    import uvicorn
+    from fastapi import FastAPI
+
+    import httpx
+
+    from unionml_amd.models.mlp import model
+
+    use_gpu = torch.cuda.is_available()
+
+    # train on synthetic digits-shaped data (no network for datasets)
+    model.artifact = None
+    model.train(
+        trainer_kwargs={"epochs": 5, "lr": 3e-3},
+        loader_kwargs=None,
+        n=4096,
+        synthetic=True,
+    )
+
+    app = FastAPI()
+    model.serve(app, batch=not args.no_batch, max_batch_size=64, max_delay_ms=0.2)
+
+    port = args.port or _free_port()
+    config = uvicorn.Config(app, host="127.0.0.1", port=port, log_level="warning")
+    server = uvicorn.Server(config)
+    thread = threading.Thread(target=server.run, daemon=True)
+    thread.start()
+    deadline = time.monotonic() + 30
+    url = f"http://127.0.0.1:{port}"
+    while time.monotonic() < deadline:
+        try:
+            if httpx.get(f"{url}/health", timeout=1.0).status_code == 200:
+                break
+        except httpx.HTTPError:
+            time.sleep(0.2)
+    else:
+        print("server never became healthy", file=sys.stderr)
+        sys.exit(1)
+
+    one_row = [{f"p{i}": float((i * 7) % 16) for i in range(64)}]
+
+    # --- sequential latency -------------------------------------------------
+    with httpx.Client(base_url=url, timeout=10.0) as client:
+        for _ in range(args.warmup):
+            client.post("/predict", json={"features": one_row})
+        lat = []
+        for _ in range(args.requests):
+            t0 = time.perf_counter()
+            r = client.post("/predict", json={"features": one_row})
+            lat.append((time.perf_counter() - t0) * 1000.0)
+            assert r.status_code == 200, r.text
+
+    # --- concurrent load ----------------------------------------------------
+    async def load():
+        async with httpx.AsyncClient(base_url=url, timeout=30.0) as client:
+
+            async def worker(n):
+                times = []
+                for _ in range(n):
+                    t0 = time.perf_counter()
+                    r = await client.post("/predict", json={"features": one_row})
+                    times.append((time.perf_counter() - t0) * 1000.0)
+                    assert r.status_code == 200
+                return times
+
+            per = max(1, args.requests // args.concurrency)
+            out = await asyncio.gather(*(worker(per) for _ in range(args.concurrency)))
+            return [t for ts in out for t in ts]
+
+    t0 = time.perf_counter()
+    conc_lat = asyncio.run(load())
+    conc_wall = time.perf_counter() - t0
+
+    # --- raw in-process graph replay (no HTTP) ------------------------------
+    raw_p50 = None
+    if use_gpu:
+        import numpy as np
+
+        from unionml_amd.serving.graph_runner import TabularGraphRunner
+
+        runner = TabularGraphRunner(model.artifact.model_object, max_batch_size=64)
+        x = np.asarray([[float((i * 7) % 16) for i in range(64)]], dtype=np.float32)
+        for _ in range(50):
+            runner(x)
+        raw = []
+        for _ in range(500):
+            t0 = time.perf_counter()
+            runner(x)
+            raw.append((time.perf_counter() - t0) * 1000.0)
+        raw_p50 = pct(raw, 0.50)
+
+    server.should_exit = True
+    thread.join(timeout=10)
+
+    print(
+        json.dumps(
+            {
+                "metric": "predict_p50_latency_ms",
+                "value": pct(lat, 0.50),
+                "unit": "ms",
+                "n_gpus": 1 if use_gpu else 0,
+                "steps": args.requests,
+                "warmup": args.warmup,
+                "higher_is_better": False,
+                "vs_baseline": None,
+                "dtype": "bf16" if use_gpu else "fp32",
+                "data": "synthetic",
+                "config": {
+                    "model": "digits_mlp_64x32x10",
+                    "batcher": not args.no_batch,
+                    "sequential": {
+                        "p50_ms": pct(lat, 0.50),
+                        "p90_ms": pct(lat, 0.90),
+                        "p99_ms": pct(lat, 0.99),
+                        "mean_ms": statistics.fmean(lat),
+                    },
+                    "concurrent": {
+                        "clients": args.concurrency,
+                        "p50_ms": pct(conc_lat, 0.50),
+                        "p99_ms": pct(conc_lat, 0.99),
+                        "requests_per_s": len(conc_lat) / conc_wall,
+                    },
+                    "raw_graph_replay_p50_ms": raw_p50,
+                },
+            }
+        )
+    )
+
+
+if __name__ == "__main__":
+    main()
