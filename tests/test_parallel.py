@@ -126,3 +126,39 @@ def test_tabular_dp2_matches_single():
     assert torch.allclose(single_master, dp_master, rtol=1e-4, atol=1e-6), (
         (single_master - dp_master).abs().max()
     )
+
+
+def _broadcast_worker(rank, world, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        # DIFFERENT init per rank; reducer must broadcast rank 0's
+        # params+buffers so all ranks start identical
+        torch.manual_seed(1000 + rank)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(6, 12), torch.nn.BatchNorm1d(12), torch.nn.Linear(12, 3)
+        )
+        GradientAllReducer(model, bucket_mb=0.0001)
+        flat = torch.cat(
+            [t.detach().reshape(-1).float() for t in list(model.parameters()) + list(model.buffers())]
+        )
+        gathered = [torch.zeros_like(flat) for _ in range(world)]
+        dist.all_gather(gathered, flat)
+        for g in gathered:
+            assert torch.equal(g, gathered[0]), "param broadcast failed"
+        with open(os.path.join(out_dir, f"ok_{rank}"), "w") as f:
+            f.write("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_reducer_broadcasts_initial_params(tmp_path):
+    from unionml_amd.parallel.launch import _free_port
+
+    port = _free_port()
+    mp.start_processes(
+        _broadcast_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True,
+        start_method="spawn",
+    )
+    assert (tmp_path / "ok_0").exists() and (tmp_path / "ok_1").exists()

@@ -92,12 +92,20 @@ class GradientAllReducer:
         module: torch.nn.Module,
         bucket_mb: float = DEFAULT_BUCKET_MB,
         process_group=None,
+        sync_params: bool = True,
     ):
         self.module = module
         self.group = process_group
         self.world = get_world_size()
         params = [p for p in module.parameters() if p.requires_grad]
         device = params[0].device if params else torch.device("cpu")
+
+        # ranks must start from identical weights: broadcast rank 0's
+        # parameters and buffers (BN running stats) once at attach time
+        if sync_params and self.world > 1 and distributed_is_active():
+            with torch.no_grad():
+                for t in list(module.parameters()) + list(module.buffers()):
+                    dist.broadcast(t.data, src=0, group=self.group)
 
         # reverse registration order ~ the order backward produces grads
         self.buckets: List[_Bucket] = []

@@ -53,6 +53,15 @@ class DynamicBatcher:
         self._graphed = None  # set lazily on first GPU batch
 
     def start(self):
+        # build the hipGraph runner (and capture all buckets) up front so
+        # no request pays the capture cost
+        predictor = self.model._predictor
+        factory = getattr(predictor, "__unionml_graphed__", None)
+        if factory is not None and self.model.artifact is not None:
+            try:
+                self._graphed = factory(self.model.artifact.model_object, self.max_batch_size)
+            except Exception:
+                logger.exception("graph runner init failed; serving eagerly")
         self._thread = threading.Thread(target=self._worker, daemon=True, name="unionml-batcher")
         self._thread.start()
 

@@ -20,7 +20,7 @@ from unionml_amd.serving.batcher import bucket_for
 class TabularGraphRunner:
     """Bucketed hipGraph replay around TabularMLP's fused predict kernel."""
 
-    def __init__(self, mlp, max_batch_size: int = 64):
+    def __init__(self, mlp, max_batch_size: int = 64, precapture: bool = True):
         from unionml_amd.ops.tabular import TabularMLP
 
         assert isinstance(mlp, TabularMLP)
@@ -32,6 +32,15 @@ class TabularGraphRunner:
             self._pinned = torch.empty(
                 max_batch_size, 64, dtype=torch.float32, pin_memory=True
             )
+            if precapture:
+                # capture every bucket up front so a first request of a
+                # new size never pays ~100ms of graph capture (observed
+                # as p99 spikes under concurrent load)
+                b = 1
+                while b < max_batch_size:
+                    self._get_bucket(b)
+                    b <<= 1
+                self._get_bucket(max_batch_size)
 
     def _get_bucket(self, b: int):
         from unionml_amd.ops import hip_ext
