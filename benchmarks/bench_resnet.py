@@ -74,9 +74,12 @@ def main():
 
     B, M = args.batch, args.minibatches
     gen = torch.Generator(device="cpu").manual_seed(100 + rank)
-    X = torch.rand(M, B, 3, args.image_size, args.image_size, generator=gen).to(device)
+    X = [
+        torch.rand(B, 3, args.image_size, args.image_size, generator=gen).to(device)
+        for _ in range(M)
+    ]
     if use_gpu:
-        X = X.to(memory_format=torch.channels_last)
+        X = [x.to(memory_format=torch.channels_last) for x in X]
     y = torch.randint(0, args.classes, (M, B), generator=gen).to(device)
 
     amp_dtype = torch.bfloat16

@@ -71,7 +71,7 @@ def main():
     )
 
     app = FastAPI()
-    model.serve(app, batch=not args.no_batch, max_batch_size=64, max_delay_ms=0.2)
+    model.serve(app, batch=not args.no_batch, max_batch_size=64, max_delay_ms=0.0)
 
     port = args.port or _free_port()
     config = uvicorn.Config(app, host="127.0.0.1", port=port, log_level="warning")
@@ -142,6 +142,18 @@ def main():
             raw.append((time.perf_counter() - t0) * 1000.0)
         raw_p50 = pct(raw, 0.50)
 
+    batcher_stats = None
+    try:
+        # dig the batcher out of the serving closure for server-side stats
+        for route in app.router.on_startup:
+            cells = getattr(route, "__closure__", None) or []
+            for c in cells:
+                v = c.cell_contents
+                if isinstance(v, dict) and v.get("batcher") is not None:
+                    batcher_stats = v["batcher"].stats()
+    except Exception:
+        pass
+
     server.should_exit = True
     thread.join(timeout=10)
 
@@ -174,6 +186,7 @@ def main():
                         "requests_per_s": len(conc_lat) / conc_wall,
                     },
                     "raw_graph_replay_p50_ms": raw_p50,
+                    "batcher_server_side": batcher_stats,
                 },
             }
         )

@@ -23,7 +23,7 @@ def serving_app(
     model_version: str = "latest",
     batch: bool = False,
     max_batch_size: int = 64,
-    max_delay_ms: float = 2.0,
+    max_delay_ms: float = 0.0,
 ):
     from fastapi import Body, HTTPException
 

@@ -588,7 +588,7 @@ class Model(TrackedInstance):
         model_version: str = "latest",
         batch: bool = False,
         max_batch_size: int = 64,
-        max_delay_ms: float = 2.0,
+        max_delay_ms: float = 0.0,
     ):
         """Attach serving routes for this model to a FastAPI app
         (reference: model.py:771-784). ``batch=True`` enables the

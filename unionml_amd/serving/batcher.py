@@ -29,6 +29,7 @@ class _Request:
     n_rows: int
     future: "asyncio.Future"
     loop: "asyncio.AbstractEventLoop"
+    t_submit: float = 0.0
 
 
 def bucket_for(n: int, max_batch: int) -> int:
@@ -42,7 +43,7 @@ def bucket_for(n: int, max_batch: int) -> int:
 class DynamicBatcher:
     """Thread-backed micro-batcher feeding a single predictor."""
 
-    def __init__(self, model, max_batch_size: int = 64, max_delay_ms: float = 2.0):
+    def __init__(self, model, max_batch_size: int = 64, max_delay_ms: float = 0.0):
         self.model = model
         self.max_batch_size = max_batch_size
         self.max_delay_s = max_delay_ms / 1000.0
@@ -51,6 +52,9 @@ class DynamicBatcher:
         self._stop = False
         self._thread: Optional[threading.Thread] = None
         self._graphed = None  # set lazily on first GPU batch
+        self._n_batches = 0
+        self._batch_rows: List[int] = []
+        self._done_latency_ms: List[float] = []
 
     def start(self):
         # build the hipGraph runner (and capture all buckets) up front so
@@ -76,11 +80,29 @@ class DynamicBatcher:
         loop = asyncio.get_running_loop()
         future = loop.create_future()
         n_rows = len(features_raw) if hasattr(features_raw, "__len__") else 1
-        req = _Request(features_raw=features_raw, n_rows=n_rows, future=future, loop=loop)
+        req = _Request(
+            features_raw=features_raw,
+            n_rows=n_rows,
+            future=future,
+            loop=loop,
+            t_submit=time.monotonic(),
+        )
         with self._cv:
             self._queue.append(req)
             self._cv.notify()
         return await future
+
+    def stats(self) -> dict:
+        """Server-side batching stats (for bench/diagnostics)."""
+        lat = sorted(self._done_latency_ms)
+        pick = lambda q: lat[min(len(lat) - 1, int(q * len(lat)))] if lat else None  # noqa: E731
+        return {
+            "batches": self._n_batches,
+            "requests": len(lat),
+            "rows_per_batch": (sum(self._batch_rows) / max(1, len(self._batch_rows))),
+            "server_p50_ms": pick(0.50),
+            "server_p99_ms": pick(0.99),
+        }
 
     # ------------------------------------------------------------------
 
@@ -92,15 +114,21 @@ class DynamicBatcher:
                 self._cv.wait(timeout=0.1)
             if self._stop and not self._queue:
                 return []
-            deadline = time.monotonic() + self.max_delay_s
-            while (
-                sum(r.n_rows for r in self._queue) < self.max_batch_size
-                and time.monotonic() < deadline
-                and not self._stop
-            ):
-                remaining = deadline - time.monotonic()
-                if remaining > 0:
-                    self._cv.wait(timeout=remaining)
+            # max_delay 0 = adaptive batching: take whatever is queued
+            # NOW — under load, requests naturally accumulate while the
+            # previous batch is in flight, so coalescing emerges without
+            # adding latency (and without the GIL-thrashing micro-wait
+            # loop that starves the server's event loop)
+            if self.max_delay_s > 0:
+                deadline = time.monotonic() + self.max_delay_s
+                while (
+                    sum(r.n_rows for r in self._queue) < self.max_batch_size
+                    and time.monotonic() < deadline
+                    and not self._stop
+                ):
+                    remaining = deadline - time.monotonic()
+                    if remaining > 0:
+                        self._cv.wait(timeout=remaining)
             batch, rows = [], 0
             while self._queue and rows + self._queue[0].n_rows <= self.max_batch_size:
                 req = self._queue.pop(0)
@@ -163,7 +191,11 @@ class DynamicBatcher:
                 continue
             try:
                 results = self._predict_batch([r.features_raw for r in batch])
+                now = time.monotonic()
+                self._n_batches += 1
+                self._batch_rows.append(sum(r.n_rows for r in batch))
                 for req, res in zip(batch, results):
+                    self._done_latency_ms.append((now - req.t_submit) * 1000.0)
                     req.loop.call_soon_threadsafe(req.future.set_result, _jsonable(res))
             except Exception as exc:
                 logger.exception("batched prediction failed")
