@@ -47,6 +47,10 @@ def main():
     device = torch.device("cuda", local_rank) if use_gpu else torch.device("cpu")
     if use_gpu:
         torch.cuda.set_device(device)
+        # let MIOpen autotune each conv shape during warmup — without
+        # this the default find path picks a bwd-weight kernel that is
+        # ~50x slower on the 7x7 stem (measured: profiles/)
+        torch.backends.cudnn.benchmark = True
 
     dist = None
     if world > 1:

@@ -161,6 +161,7 @@ def trainer(
     net = net.to(device)
     if device.type == "cuda":
         net = net.to(memory_format=torch.channels_last)
+        torch.backends.cudnn.benchmark = True  # MIOpen conv autotuning
     reducer = maybe_wrap(net)
     opt = torch.optim.Adam(net.parameters(), lr=lr, foreach=True)
     X = torch.from_numpy(np.ascontiguousarray(images))
