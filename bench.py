@@ -96,8 +96,8 @@ def main():
     from unionml_amd.ops.reference import NPARAM
 
     B, M = args.batch, args.minibatches
-    torch.manual_seed(1234 + rank)
-    clf = TabularMLP(device=device, seed=rank)
+    torch.manual_seed(1234 + rank)  # per-rank DATA; weights identical (seed=0)
+    clf = TabularMLP(device=device, seed=0)
 
     # synthetic digits-shaped data, staged bf16-resident in HBM
     X = torch.rand(B * M, 64, device=clf.device) * 16.0
