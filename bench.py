@@ -37,6 +37,12 @@ def parse_args():
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--no-graph", action="store_true")
     p.add_argument(
+        "--graph-steps",
+        type=int,
+        default=16,
+        help="optimizer steps captured per hipGraph (amortizes replay launch)",
+    )
+    p.add_argument(
         "--engine",
         choices=["auto", "fused", "persistent", "stepwise"],
         default="auto",
@@ -170,26 +176,38 @@ def main():
     for i in range(3):
         eager_step((i % M) * B)
 
+    # capture G optimizer steps per graph (G | M), so one replay advances
+    # G steps — all work identical, launch overhead amortized G-fold
+    G = max(1, args.graph_steps)
+    while M % G:
+        G -= 1
     graphs = None
     if use_gpu and not args.no_graph:
         try:
             graphs = []
-            for mb in range(M):
+            for chunk in range(M // G):
                 g = torch.cuda.CUDAGraph()
                 with torch.cuda.graph(g):
-                    eager_step(mb * B)
+                    for j in range(G):
+                        eager_step((chunk * G + j) * B)
                 graphs.append(g)
         except RuntimeError as exc:
             print(f"[bench] graph capture unavailable ({exc}); eager stepping",
                   file=sys.stderr)
             graphs = None
 
-    def step(k):
-        mb = k % M
-        if graphs is not None:
-            graphs[mb].replay()
-        else:
-            eager_step(mb * B)
+    def run_steps(k0: int, nsteps: int):
+        """Advance exactly ``nsteps`` optimizer steps from global step k0:
+        G-aligned stretches replay whole-chunk graphs, edges run eagerly."""
+        k = k0
+        end = k0 + nsteps
+        while k < end:
+            if graphs is not None and k % G == 0 and k + G <= end:
+                graphs[(k // G) % (M // G)].replay()
+                k += G
+            else:
+                eager_step((k % M) * B)
+                k += 1
 
     def barrier_sync():
         if dist is not None:
@@ -197,12 +215,10 @@ def main():
         if use_gpu:
             torch.cuda.synchronize()
 
-    for k in range(args.warmup):
-        step(k)
+    run_steps(0, args.warmup)
     barrier_sync()
     t0 = time.perf_counter()
-    for k in range(args.steps):
-        step(k)
+    run_steps(args.warmup, args.steps)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
@@ -216,7 +232,7 @@ def main():
     assert loss == loss and loss < 1e6, f"training diverged: loss={loss}"
 
     if rank == 0:
-        engine_name = f"{engine}+{'hipgraph' if graphs is not None else 'eager'}"
+        engine_name = f"{engine}+{f'hipgraph{G}' if graphs is not None else 'eager'}"
         print(json.dumps(_result(args, n_gpus, B, elapsed, engine_name, loss)))
 
     if dist is not None:
