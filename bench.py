@@ -5,9 +5,10 @@ BASELINE.json metric: "train samples/sec + /predict p50 latency, MLP
 digits-shape, 1/2/4/8 MI355X". This measures the training leg on
 synthetic digits-shaped data (64 features, 10 classes, random-init
 weights — no network for datasets) with the CDNA4 fused hot path:
-one optimizer step = zero-grads + fused fwd/bwd MFMA kernel
-[+ RCCL gradient all-reduce under DP] + fused Adam, replayed from
-per-minibatch hipGraphs.
+one optimizer step = the fully-fused fwd/bwd/reduce/Adam MFMA kernel
+(or, under DP, fused fwd/bwd kernel + RCCL gradient all-reduce + fused
+Adam), with 16 optimizer steps captured per hipGraph so replay launch
+overhead is amortized (--graph-steps).
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W              # single GPU

@@ -1,0 +1,66 @@
+"""Driver-contract tests for the benchmark scripts: they must run on
+CPU with default-compatible flags and print ONE JSON line with the
+agreed schema (the round driver invokes them verbatim)."""
+
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).parent.parent
+
+REQUIRED_FIELDS = {
+    "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+    "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config",
+}
+
+
+def _run(cmd, timeout=240):
+    env = dict(os.environ, PYTHONPATH=str(REPO) + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    proc = subprocess.run(
+        cmd, cwd=REPO, capture_output=True, text=True, timeout=timeout, env=env
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    line = proc.stdout.strip().splitlines()[-1]
+    return json.loads(line)
+
+
+@pytest.mark.timeout(300)
+def test_bench_default_contract_cpu():
+    d = _run([sys.executable, "bench.py", "--steps", "6", "--warmup", "2",
+              "--batch", "64", "--minibatches", "4"])
+    assert REQUIRED_FIELDS <= set(d)
+    assert d["metric"] == "train_samples_per_sec"
+    assert d["n_gpus"] == 1 and d["steps"] == 6 and d["warmup"] == 2
+    assert d["higher_is_better"] is True and d["scaling"] == "weak"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["config"]["model"] == "digits_mlp_64x32x10"
+    assert d["config"]["global_batch"] == 64
+    assert d["config"]["parallelism"] == "dp1"
+    assert d["data"] == "synthetic"
+
+
+@pytest.mark.timeout(300)
+def test_bench_world2_gloo_cpu():
+    """The exact launch shape the driver uses for N>1, on gloo/CPU."""
+    d = _run([
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+        "--master-port", "29590", "bench.py", "--gpus", "2", "--steps", "4",
+        "--warmup", "1", "--batch", "32", "--minibatches", "4",
+    ])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 64
+
+
+@pytest.mark.timeout(300)
+def test_bench_resnet_contract_cpu():
+    d = _run([sys.executable, "benchmarks/bench_resnet.py", "--steps", "2",
+              "--warmup", "1", "--batch", "2", "--image-size", "32",
+              "--classes", "4"])
+    assert REQUIRED_FIELDS <= set(d)
+    assert d["config"]["model"] == "resnet18"

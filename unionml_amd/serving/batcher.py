@@ -17,7 +17,7 @@ predictor — one code path, two execution substrates.
 import asyncio
 import threading
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, List, Optional
 
 from unionml_amd._logging import logger
