@@ -532,23 +532,32 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory"); // keep: ROCm may drop it
     __hip_atomic_store((unsigned*)(slab + NPARAM + 1), epoch, __ATOMIC_RELAXED,
                        __HIP_MEMORY_SCOPE_AGENT);
-    // relaxed sweep of every slab's tag word, bounded spin (G16: never
-    // poll with an acquire)
+  }
+  __syncthreads();                                   // tag stored before polls
+  if (wave == 0) {
+    // relaxed PARALLEL sweep: lane j polls tags j, j+64, ... (G16: never
+    // poll with an acquire). One vector gather per poll round — one
+    // memory round trip after the last publish, independent of n_wg
+    // (the serial lane-0 sweep this replaces cost ~500 cyc per WG and
+    // dominated the handshake at large grids).
     unsigned spins = 0;
     bool ok = true;
-    for (int w = 0; w < n_wg;) {
-      const unsigned tag = __hip_atomic_load(
-          (const unsigned*)(slabs + (long long)w * SLAB + NPARAM + 1),
-          __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-      if (tag == epoch) {
-        ++w;
-      } else {
+    for (int w = l; w < n_wg; w += 64) {
+      for (;;) {
+        const unsigned tag = __hip_atomic_load(
+            (const unsigned*)(slabs + (long long)w * SLAB + NPARAM + 1),
+            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        if (tag == epoch) break;
         __builtin_amdgcn_s_sleep(2);
         if (++spins > 100000000u) { ok = false; break; }
       }
+      if (!ok) break;
     }
-    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-    lossu[1] = ok ? 0u : 1u;
+    const unsigned long long bad = __ballot(!ok);
+    if (l == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      lossu[1] = bad ? 1u : 0u;
+    }
   }
   __syncthreads();
   if (lossu[1] != 0u) {             // timed out: poison the loss, keep going
@@ -565,15 +574,20 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   const int hi = min(lo + span, NPARAM + 1);
   for (int i = lo + tid; i < hi; i += BLOCK) {
     float g0 = 0.f, g1 = 0.f, g2 = 0.f, g3 = 0.f;
+    float g4 = 0.f, g5 = 0.f, g6 = 0.f, g7 = 0.f;
     int w = 0;
-    for (; w + 4 <= n_wg; w += 4) {      // 4 independent load chains
+    for (; w + 8 <= n_wg; w += 8) {      // 8 independent load chains
       g0 += slabs[(long long)w * SLAB + i];
       g1 += slabs[(long long)(w + 1) * SLAB + i];
       g2 += slabs[(long long)(w + 2) * SLAB + i];
       g3 += slabs[(long long)(w + 3) * SLAB + i];
+      g4 += slabs[(long long)(w + 4) * SLAB + i];
+      g5 += slabs[(long long)(w + 5) * SLAB + i];
+      g6 += slabs[(long long)(w + 6) * SLAB + i];
+      g7 += slabs[(long long)(w + 7) * SLAB + i];
     }
     for (; w < n_wg; ++w) g0 += slabs[(long long)w * SLAB + i];
-    const float g = (g0 + g1) + (g2 + g3);
+    const float g = ((g0 + g1) + (g2 + g3)) + ((g4 + g5) + (g6 + g7));
     if (i == NPARAM) {
       *loss_out = g;
       continue;
