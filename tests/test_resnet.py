@@ -70,3 +70,25 @@ def test_resnet_dp2_trains_on_cpu():
     )
     assert set(metrics) == {"train", "test"}
     assert sum(p.numel() for p in net.parameters()) > 0
+
+
+def test_module_graph_runner_cpu_fallback():
+    """On CPU the ModuleGraphRunner runs a plain no-grad forward and must
+    match the predictor exactly."""
+    from unionml_amd.models.resnet import ResNet18
+    from unionml_amd.serving.graph_runner import ModuleGraphRunner
+
+    torch.manual_seed(3)
+    net = ResNet18(**TINY)
+    runner = ModuleGraphRunner(net, max_batch_size=8, postprocess="argmax")
+    x = np.random.RandomState(1).rand(5, 3, 32, 32).astype(np.float32)
+    out = runner(x)
+    with torch.no_grad():
+        direct = net.eval()(torch.from_numpy(x)).argmax(dim=1).numpy()
+    assert (out == direct).all()
+
+    logits_runner = ModuleGraphRunner(net, max_batch_size=8, postprocess=None)
+    logits = logits_runner(x)
+    with torch.no_grad():
+        ref = net.eval()(torch.from_numpy(x)).numpy()
+    assert np.allclose(logits, ref, atol=1e-6)

@@ -23,6 +23,7 @@ import torch.nn.functional as F
 
 from unionml_amd import Dataset, Model
 from unionml_amd.parallel import maybe_wrap
+from unionml_amd.serving.graph_runner import graphed
 
 
 class BasicBlock(nn.Module):
@@ -188,7 +189,16 @@ def trainer(
     return net
 
 
+def _graph_runner(net, max_batch):
+    from unionml_amd.serving.graph_runner import ModuleGraphRunner
+
+    return ModuleGraphRunner(
+        net, max_batch_size=max_batch, channels_last=True, postprocess="argmax"
+    )
+
+
 @model.predictor
+@graphed(_graph_runner)
 def predictor(net: ResNet18, images: np.ndarray) -> np.ndarray:
     device = next(net.parameters()).device
     net.eval()
