@@ -195,7 +195,10 @@ class ModuleGraphRunner:
             g.replay()
             outs.append(out_static[:chunk].cpu())
             off += chunk
-        return torch.cat(outs).numpy()
+        res = torch.cat(outs)
+        if res.dtype == torch.bfloat16:
+            res = res.float()
+        return res.numpy()
 
 
 def graphed(factory: Callable):
