@@ -137,11 +137,20 @@ def main():
                 assert ok
 
         else:
+            # DP path: fused fwd/bwd + in-kernel slab reduction writing the
+            # summed grads (no global atomics, no zeroing kernel), then the
+            # RCCL all-reduce, then fused Adam
+            clf._ensure_slabs((B + 127) // 128)
 
             def eager_step(off):
-                clf.grads.zero_()
-                ext.mlp_step(Xbf[off : off + B], y[off : off + B], clf.W1bf,
-                             clf.W2bf, clf.master, clf.grads, invBtot)
+                ok = ext.mlp_step_fused(
+                    Xbf[off : off + B], y[off : off + B], clf.W1bf, clf.W2bf,
+                    clf.master, clf.bfmirror, clf.m, clf.v, clf.t_dev,
+                    clf.slabs, clf.counter, loss_out, invBtot,
+                    args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                    grads_out=clf.grads,
+                )
+                assert ok
                 if dist is not None:
                     dist.all_reduce(clf.grads)
                 ext.adam_step(clf.master, clf.bfmirror, clf.grads, clf.m, clf.v,

@@ -257,3 +257,41 @@ def test_mlp_app_end_to_end_gpu(ext, dev):
     feats = [{f"p{i}": float(i % 16) for i in range(64)}]
     preds = model.predict(features=feats)
     assert len(preds) == 1
+
+
+def test_step_fused_reduce_only_matches_mlp_step(ext, dev):
+    """grads_out mode: same summed grads as the atomic-accumulation
+    mlp_step kernel, with Adam state untouched."""
+    from unionml_amd.ops import reference as ref
+    from unionml_amd.ops.tabular import TabularMLP
+
+    torch.manual_seed(31)
+    B = 384  # 3 workgroups
+    Xbf = (torch.randn(B, 64) * 1.3 - 0.1).bfloat16().to(dev)
+    y = torch.randint(0, 10, (B,), dtype=torch.int32, device=dev)
+
+    a = TabularMLP(device=dev, seed=8)
+    a.grads.zero_()
+    ext.mlp_step(Xbf, y, a.W1bf, a.W2bf, a.master, a.grads, 1.0 / B)
+
+    b = TabularMLP(device=dev, seed=8)
+    b._ensure_slabs((B + 127) // 128)
+    master0 = b.master.clone()
+    loss_out = torch.zeros(1, device=dev)
+    for _ in range(2):  # twice: launch-epoch counter must advance per call
+        ok = ext.mlp_step_fused(
+            Xbf, y, b.W1bf, b.W2bf, b.master, b.bfmirror, b.m, b.v, b.t_dev,
+            b.slabs, b.counter, loss_out, 1.0 / B, 1e-3, 0.9, 0.999, 1e-8,
+            grads_out=b.grads,
+        )
+        assert ok
+    torch.cuda.synchronize()
+
+    # Adam state untouched in reduce-only mode
+    assert int(b.t_dev.item()) == 0
+    assert torch.equal(b.master, master0)
+    assert b.m.abs().max().item() == 0.0
+    # summed grads (incl. loss at NPARAM) match the atomic path
+    err = (a.grads[: ref.NPARAM + 1] - b.grads[: ref.NPARAM + 1]).abs().max().item()
+    scale = a.grads.abs().max().item()
+    assert err < max(1e-5 * scale, 1e-6), (err, scale)

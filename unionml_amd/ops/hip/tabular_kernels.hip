@@ -462,10 +462,14 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
                       float* __restrict__ m, float* __restrict__ v,
                       int* __restrict__ t_dev,
                       float* __restrict__ slabs,    // [n_wg][SLAB]
-                      unsigned* __restrict__ counter,  // MONOTONIC ticket counter
+                      unsigned* __restrict__ counter,  // launch-epoch counter (reduce-only mode)
                       float* __restrict__ loss_out,
                       float invBtot, float lr, float beta1, float beta2,
-                      float eps) {
+                      float eps,
+                      float* __restrict__ grads_out) { // non-null: write summed
+                                                       // grads (+loss) and skip
+                                                       // Adam — the DP path's
+                                                       // pre-collective kernel
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const Lds L = carve(smem);
   // L.loss slots: [0] loss accum, [1] (bits) poll base, [2] t_pre, [3] unused
@@ -482,10 +486,12 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   if (tid < CPAD) L.db2[tid] = 0.f;
   if (tid == 0) {
     L.loss[0] = 0.f;
-    // read the step counter BEFORE any workgroup of this launch can have
+    // read the epoch source BEFORE any workgroup of this launch can have
     // advanced it (the writer only writes after every WG has published,
-    // i.e. after every WG has passed this point)
-    L.loss[2] = (float)(*t_dev);
+    // i.e. after every WG has passed this point). Adam mode: the Adam
+    // step counter t_dev; reduce-only mode: the dedicated launch counter
+    // (t_dev is advanced by the separate adam_step kernel there).
+    lossu[2] = grads_out ? *counter : (unsigned)(*t_dev);
   }
   zero_dl_pad(L);
   load_weight_images(L, W1bf, W2bf);
@@ -524,7 +530,7 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   // unique per launch (exactly one WG advances t_dev per launch), so the
   // scheme is robust to any grid size, graph replay, and engine mixing;
   // no counter, no reset. ----------------------------------------------------
-  const unsigned epoch = (unsigned)L.loss[2] + 1u;
+  const unsigned epoch = lossu[2] + 1u;
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");   // EVERY storing wave drains
   __syncthreads();
   if (tid == 0) {
@@ -566,7 +572,7 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   }
 
   // ---- every WG reduces its own param stripe + applies Adam ----------------
-  const float t_new = L.loss[2] + 1.f;
+  const float t_new = (float)(lossu[2] + 1u);
   const float corr1 = fast_rcp(1.f - __powf(beta1, t_new));
   const float corr2 = fast_rcp(1.f - __powf(beta2, t_new));
   const int span = (NPARAM + 1 + n_wg - 1) / n_wg;
@@ -588,6 +594,10 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
     }
     for (; w < n_wg; ++w) g0 += slabs[(long long)w * SLAB + i];
     const float g = ((g0 + g1) + (g2 + g3)) + ((g4 + g5) + (g6 + g7));
+    if (grads_out) {            // reduce-only: hand the summed grads (+loss
+      grads_out[i] = g;         // at NPARAM) to the RCCL all-reduce
+      continue;
+    }
     if (i == NPARAM) {
       *loss_out = g;
       continue;
@@ -600,7 +610,10 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
     master[i] = p;
     bfmirror[i] = f2bf(p);
   }
-  if (tid == 0 && blockIdx.x == 0) *t_dev = (int)t_new;
+  if (tid == 0 && blockIdx.x == 0) {
+    if (grads_out) *counter = epoch;
+    else *t_dev = (int)t_new;
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -912,7 +925,7 @@ int launch_mlp_step_fused(const unsigned short* Xbf, const int* y, int B,
                           float* master, unsigned short* bfmirror, float* m,
                           float* v, int* t_dev, float* slabs, unsigned* counter,
                           float* loss_out, float invBtot, float lr, float beta1,
-                          float beta2, float eps, int max_slabs,
+                          float beta2, float eps, int max_slabs, float* grads_out,
                           hipStream_t stream) {
   const int blocks = (B + ROWS - 1) / ROWS;
   if (blocks > max_slabs) return -1;
@@ -923,7 +936,8 @@ int launch_mlp_step_fused(const unsigned short* Xbf, const int* y, int B,
   }
   hipLaunchKernelGGL(mlp_step_fused_kernel, dim3(blocks), dim3(BLOCK), C_IMG_TOTAL,
                      stream, Xbf, y, B, W1bf, W2bf, master, bfmirror, m, v, t_dev,
-                     slabs, counter, loss_out, invBtot, lr, beta1, beta2, eps);
+                     slabs, counter, loss_out, invBtot, lr, beta1, beta2, eps,
+                     grads_out);
   return 0;
 }
 

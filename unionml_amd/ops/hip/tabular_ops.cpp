@@ -19,7 +19,7 @@ int launch_mlp_train_steps(const unsigned short*, const int*, long long, int, in
 int launch_mlp_step_fused(const unsigned short*, const int*, int,
                           const unsigned short*, const unsigned short*, float*,
                           unsigned short*, float*, float*, int*, float*, unsigned*,
-                          float*, float, float, float, float, float, int,
+                          float*, float, float, float, float, float, int, float*,
                           hipStream_t);
 void launch_mlp_predict(const float*, int, const float*, const float*,
                         const unsigned short*, const unsigned short*, const float*,
@@ -93,7 +93,8 @@ bool mlp_step_fused(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
                     torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
                     torch::Tensor slabs, torch::Tensor counter,
                     torch::Tensor loss_out, double invBtot, double lr,
-                    double beta1, double beta2, double eps) {
+                    double beta1, double beta2, double eps,
+                    c10::optional<torch::Tensor> grads_out = c10::nullopt) {
   check(Xbf, torch::kBFloat16, "Xbf");
   check(y, torch::kInt32, "y");
   check(master, torch::kFloat32, "master");
@@ -103,13 +104,19 @@ bool mlp_step_fused(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
   check(loss_out, torch::kFloat32, "loss_out");
   TORCH_CHECK(Xbf.size(1) == 64, "IN must be 64");
   TORCH_CHECK(slabs.dim() == 2 && slabs.size(1) == 2624, "slabs must be [n][2624]");
+  float* grads_ptr = nullptr;
+  if (grads_out.has_value()) {
+    check(*grads_out, torch::kFloat32, "grads_out");
+    TORCH_CHECK(grads_out->numel() >= 2609, "grads_out must hold params + loss");
+    grads_ptr = grads_out->data_ptr<float>();
+  }
   const int rc = launch_mlp_step_fused(
       bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0), bf16_ptr(W1bf),
       bf16_ptr(W2bf), master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
       m.data_ptr<float>(), v.data_ptr<float>(), t_dev.data_ptr<int>(),
       slabs.data_ptr<float>(), (unsigned*)counter.data_ptr(),
       loss_out.data_ptr<float>(), (float)invBtot, (float)lr, (float)beta1,
-      (float)beta2, (float)eps, (int)slabs.size(0), current_stream());
+      (float)beta2, (float)eps, (int)slabs.size(0), grads_ptr, current_stream());
   TORCH_CHECK(rc != -2, "mlp_step_fused: hipFuncSetAttribute(LDS) failed");
   return rc == 0;
 }
@@ -174,7 +181,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("standardize_apply", &standardize_apply, "(x-mean)*invstd -> bf16 (CDNA4)");
   m.def("mlp_step", &mlp_step, "fused MLP fwd+bwd step (CDNA4 MFMA)");
   m.def("mlp_step_fused", &mlp_step_fused,
-        "fully-fused step: fwd+bwd + cross-WG slab reduction + Adam, one launch");
+        "fully-fused step: fwd+bwd + cross-WG slab reduction + Adam in one "
+        "launch; with grads_out, reduce-only (the DP pre-collective kernel)",
+        py::arg("Xbf"), py::arg("y"), py::arg("W1bf"), py::arg("W2bf"),
+        py::arg("master"), py::arg("bfmirror"), py::arg("m"), py::arg("v"),
+        py::arg("t_dev"), py::arg("slabs"), py::arg("counter"),
+        py::arg("loss_out"), py::arg("invBtot"), py::arg("lr"), py::arg("beta1"),
+        py::arg("beta2"), py::arg("eps"), py::arg("grads_out") = c10::nullopt);
   m.def("mlp_train_steps", &mlp_train_steps,
         "persistent multi-step training kernel (weights+Adam resident in LDS)");
   m.def("mlp_predict", &mlp_predict, "fused standardize+fwd+argmax (CDNA4 MFMA)");
