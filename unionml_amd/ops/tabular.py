@@ -96,10 +96,20 @@ class TabularMLP:
 
     def _step(self, Xbf: torch.Tensor, y: torch.Tensor, invBtot: float, lr: float,
               allreduce: bool):
-        self.grads.zero_()
         if self.use_hip:
-            hip_ext().mlp_step(Xbf, y, self.W1bf, self.W2bf, self.master, self.grads, invBtot)
+            # fused fwd/bwd + in-kernel slab reduction writing the summed
+            # grads (reduce-only mode: no global atomics, no zeroing pass)
+            self._ensure_slabs((Xbf.shape[0] + 127) // 128)
+            loss_out = self.grads[NPARAM : NPARAM + 1]
+            ok = hip_ext().mlp_step_fused(
+                Xbf, y, self.W1bf, self.W2bf, self.master, self.bfmirror,
+                self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
+                invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                grads_out=self.grads,
+            )
+            assert ok, "mlp_step_fused slab capacity exceeded"
         else:
+            self.grads.zero_()
             ref.mlp_step(Xbf, y, self.W1bf, self.W2bf, self.master, self.grads, invBtot)
         if allreduce:
             import torch.distributed as dist
