@@ -107,14 +107,22 @@ class Task:
         resources: Resources = DEFAULT_RESOURCES,
         cache: bool = False,
         cache_version: str = "0",
+        retries: int = 0,
+        retry_delay_s: float = 0.0,
         resolver_args: Optional[ResolverArgs] = None,
         signature: Optional[inspect.Signature] = None,
+        **extra_task_kwargs,
     ):
         self.fn = fn
         self.name = name
         self.resources = resources
         self.cache = cache
         self.cache_version = cache_version
+        self.retries = retries
+        self.retry_delay_s = retry_delay_s
+        # unknown task kwargs are kept as metadata (the reference forwards
+        # arbitrary kwargs to flytekit's task(); we stay tolerant the same way)
+        self.extra_task_kwargs = extra_task_kwargs
         self.resolver_args = resolver_args
         self._signature = signature or inspect.signature(fn)
         functools.update_wrapper(self, fn, updated=[])
@@ -131,7 +139,19 @@ class Task:
                 if hit:
                     logger.info("task %s: cache hit", self.name)
                     return value
-        out = self.fn(*args, **kwargs)
+        for attempt in range(self.retries + 1):
+            try:
+                out = self.fn(*args, **kwargs)
+                break
+            except Exception as exc:
+                if attempt >= self.retries:
+                    raise
+                logger.warning(
+                    "task %s failed (attempt %d/%d): %s — retrying",
+                    self.name, attempt + 1, self.retries + 1, exc,
+                )
+                if self.retry_delay_s > 0:
+                    time.sleep(self.retry_delay_s)
         if self.cache and key is not None:
             _GLOBAL_TASK_CACHE.put(key, out)
         return out
@@ -220,6 +240,7 @@ def inner_task(
     cache_version: str = "0",
     task_builder: Optional[str] = None,
     signature: Optional[inspect.Signature] = None,
+    **task_kwargs,
 ) -> Task:
     """Wrap a closure into a :class:`Task` named ``{owner.name}.{fn name}``
     with resolver metadata (reference: unionml/utils.py:11-60).
@@ -248,6 +269,7 @@ def inner_task(
         cache_version=cache_version,
         resolver_args=resolver,
         signature=signature,
+        **task_kwargs,
     )
     task.__unionml_object__ = owner
     return task
