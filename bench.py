@@ -97,6 +97,12 @@ def main():
 
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        if use_gpu:
+            # RCCL collectives inside hipGraph capture require the async
+            # error watchdog off (else capture aborts / the watchdog can
+            # fire mid-capture); harmless for a short bench
+            os.environ.setdefault("NCCL_ASYNC_ERROR_HANDLING", "0")
+            os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "0")
         dist.init_process_group("nccl" if use_gpu else "gloo")
 
     from unionml_amd.ops.tabular import ADAM_BETA1, ADAM_BETA2, ADAM_EPS, TabularMLP
