@@ -24,12 +24,19 @@ class PinnedStager:
         self.device = torch.device(device) if device is not None else torch.device(
             "cuda" if torch.cuda.is_available() else "cpu"
         )
-        self._pinned: dict = {}  # (dtype, key) -> host buffer, grown geometrically
+        self._pinned: dict = {}  # dtype -> host buffer, grown geometrically
+        self._inflight: dict = {}  # dtype -> event of the last async copy FROM the buffer
         self.copy_stream = (
             torch.cuda.Stream(device=self.device) if self.device.type == "cuda" else None
         )
 
     def _pinned_buffer(self, numel: int, dtype):
+        # the previous async H2D reading this buffer must have completed
+        # before the host overwrites it (wait_stream orders GPU streams,
+        # not host writes — without this, back-to-back stagings race)
+        ev = self._inflight.pop(dtype, None)
+        if ev is not None:
+            ev.synchronize()
         buf = self._pinned.get(dtype)
         if buf is None or buf.numel() < numel:
             cap = max(numel, 2 * buf.numel() if buf is not None else numel)
@@ -65,6 +72,9 @@ class PinnedStager:
         with torch.cuda.stream(stream):
             dev = torch.empty(t.shape, dtype=t.dtype, device=self.device)
             dev.reshape(-1).copy_(host, non_blocking=non_blocking)
+            ev = torch.cuda.Event()
+            ev.record(stream)
+            self._inflight[t.dtype] = ev
         if self.copy_stream is not None:
             torch.cuda.current_stream(self.device).wait_stream(self.copy_stream)
         return dev
