@@ -68,7 +68,9 @@ def test_backend_gpu_execution_and_fanout(tmp_path, monkeypatch):
     if not torch.cuda.is_available():
         pytest.skip("needs MI355X")
     backend_path = tmp_path / "backend"
-    (tmp_path / "gpu_backend_app.py").write_text(APP.format(backend_path=backend_path))
+    (tmp_path / "gpu_backend_app.py").write_text(
+        APP.replace("{backend_path}", str(backend_path))
+    )
     monkeypatch.chdir(tmp_path)
     monkeypatch.syspath_prepend(str(tmp_path))
     try:
