@@ -108,3 +108,21 @@ def test_mlp_app_save_load_roundtrip(tmp_path):
     model.artifact = None
     model.load(str(path))
     assert model.predict(features=feats) == before
+
+
+def test_tabular_mlp_pickle_roundtrip():
+    """TabularMLP must pickle CPU-portably (backend process boundary)."""
+    import pickle
+
+    import torch
+
+    from unionml_amd.ops.tabular import TabularMLP
+
+    clf = TabularMLP(device="cpu", seed=4)
+    clf.custom_attr = "kept"
+    blob = pickle.dumps(clf)
+    back = pickle.loads(blob)
+    assert back.custom_attr == "kept"
+    for k, v in clf.state_dict().items():
+        assert torch.equal(v, back.state_dict()[k]), k
+    assert back.slabs is None and back._graph is None

@@ -271,6 +271,33 @@ class TabularMLP:
 
     # -- persistence -----------------------------------------------------------
 
+    def __getstate__(self):
+        """Pickle CPU-portably: device caches (slabs, graphs) are
+        rebuilt lazily; tensors travel as CPU copies. This is what lets
+        a GPU-trained classifier cross the backend's process boundary
+        and load on any machine."""
+        d = dict(self.__dict__)
+        d["slabs"] = d["counter"] = None
+        d["_graph"] = d["_graph_key"] = None
+        d["device"] = str(self.device)
+        for k, v in list(d.items()):
+            if torch.is_tensor(v):
+                d[k] = v.detach().cpu()
+        return d
+
+    def __setstate__(self, state):
+        device = torch.device(state.pop("device"))
+        if device.type == "cuda" and not torch.cuda.is_available():
+            device = torch.device("cpu")
+        self.__dict__.update(state)
+        self.device = device
+        self.use_hip = device.type == "cuda"
+        if self.use_hip:
+            hip_ext(required=True)
+        for k, v in list(self.__dict__.items()):
+            if torch.is_tensor(v):
+                setattr(self, k, v.to(device))
+
     def state_dict(self) -> Dict[str, torch.Tensor]:
         W1, b1, W2, b2 = ref.unpack_master(self.master.cpu())
         return {

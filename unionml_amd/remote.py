@@ -306,7 +306,9 @@ class Backend:
             "model_name": manifest["model_name"],
         }
         (exec_dir / "job.json").write_text(json.dumps(job))
-        (exec_dir / "inputs.pkl").write_bytes(cloudpickle.dumps(inputs or {}))
+        from unionml_amd.utils.serialization import tensors_to_cpu
+
+        (exec_dir / "inputs.pkl").write_bytes(cloudpickle.dumps(tensors_to_cpu(inputs or {})))
 
         env = dict(os.environ)
         if n_gpus is None:

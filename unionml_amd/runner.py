@@ -62,7 +62,13 @@ def main(exec_dir: str) -> int:
         else:
             raise ValueError(f"unknown workflow {workflow!r}")
 
-        (exec_path / "outputs.pkl").write_bytes(cloudpickle.dumps(outputs))
+        from unionml_amd.utils.serialization import tensors_to_cpu
+
+        # device payloads must be CPU-portable across the process
+        # boundary (plain pickle of CUDA storages does not round-trip)
+        (exec_path / "outputs.pkl").write_bytes(
+            cloudpickle.dumps(tensors_to_cpu(outputs, _copy_modules=False))
+        )
         (exec_path / "status").write_text("SUCCEEDED")
         return 0
     except Exception:
