@@ -35,6 +35,12 @@ def parse_args():
     p.add_argument("--concurrency", type=int, default=16)
     p.add_argument("--port", type=int, default=0)
     p.add_argument("--no-batch", action="store_true")
+    p.add_argument(
+        "--workers",
+        type=int,
+        default=0,
+        help=">0: serve via `unionml-amd serve` subprocess with N uvicorn workers",
+    )
     return p.parse_args()
 
 
@@ -70,17 +76,45 @@ def main():
         synthetic=True,
     )
 
-    app = FastAPI()
-    model.serve(app, batch=not args.no_batch, max_batch_size=64, max_delay_ms=0.0)
-
     port = args.port or _free_port()
-    config = uvicorn.Config(app, host="127.0.0.1", port=port, log_level="warning")
-    server = uvicorn.Server(config)
-    thread = threading.Thread(target=server.run, daemon=True)
-    thread.start()
-    deadline = time.monotonic() + 30
     url = f"http://127.0.0.1:{port}"
+    server = thread = proc = None
+    if args.workers > 0:
+        # real multi-process serving through the CLI: save the artifact,
+        # spawn `unionml-amd serve` with N uvicorn workers (each loads
+        # the artifact and captures its own hipGraphs)
+        import subprocess
+        import tempfile
+
+        tmp = tempfile.mkdtemp(prefix="unionml_serve_bench_")
+        artifact_path = os.path.join(tmp, "model.pt")
+        model.save(artifact_path)
+        env = dict(os.environ)
+        env.pop("UNIONML_MODEL_PATH", None)
+        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        proc = subprocess.Popen(
+            [
+                sys.executable, "-m", "unionml_amd.cli", "serve",
+                "unionml_amd.models.mlp_serve:app",
+                "--model-path", artifact_path,
+                "--port", str(port), "--workers", str(args.workers),
+            ],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, env=env,
+        )
+    else:
+        app = FastAPI()
+        model.serve(app, batch=not args.no_batch, max_batch_size=64, max_delay_ms=0.0)
+        config = uvicorn.Config(app, host="127.0.0.1", port=port, log_level="warning")
+        server = uvicorn.Server(config)
+        thread = threading.Thread(target=server.run, daemon=True)
+        thread.start()
+
+    deadline = time.monotonic() + 120
     while time.monotonic() < deadline:
+        if proc is not None and proc.poll() is not None:
+            print(proc.stdout.read().decode()[-3000:], file=sys.stderr)
+            sys.exit(1)
         try:
             if httpx.get(f"{url}/health", timeout=1.0).status_code == 200:
                 break
@@ -143,19 +177,25 @@ def main():
         raw_p50 = pct(raw, 0.50)
 
     batcher_stats = None
-    try:
-        # dig the batcher out of the serving closure for server-side stats
-        for route in app.router.on_startup:
-            cells = getattr(route, "__closure__", None) or []
-            for c in cells:
-                v = c.cell_contents
-                if isinstance(v, dict) and v.get("batcher") is not None:
-                    batcher_stats = v["batcher"].stats()
-    except Exception:
-        pass
-
-    server.should_exit = True
-    thread.join(timeout=10)
+    if server is not None:
+        try:
+            # dig the batcher out of the serving closure for server-side stats
+            for route in app.router.on_startup:
+                cells = getattr(route, "__closure__", None) or []
+                for c in cells:
+                    v = c.cell_contents
+                    if isinstance(v, dict) and v.get("batcher") is not None:
+                        batcher_stats = v["batcher"].stats()
+        except Exception:
+            pass
+        server.should_exit = True
+        thread.join(timeout=10)
+    if proc is not None:
+        proc.terminate()
+        try:
+            proc.wait(timeout=10)
+        except Exception:
+            proc.kill()
 
     print(
         json.dumps(
@@ -173,6 +213,7 @@ def main():
                 "config": {
                     "model": "digits_mlp_64x32x10",
                     "batcher": not args.no_batch,
+                    "workers": args.workers or 1,
                     "sequential": {
                         "p50_ms": pct(lat, 0.50),
                         "p90_ms": pct(lat, 0.90),

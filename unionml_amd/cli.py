@@ -364,9 +364,22 @@ def serve(
     port: int = typer.Option(8000, "--port", "-p"),
     workers: int = typer.Option(1, "--workers"),
     reload: bool = typer.Option(False, "--reload"),
+    reuse_port_worker: bool = typer.Option(
+        False, "--reuse-port-worker", hidden=True,
+        help="internal: run as one SO_REUSEPORT worker of a multi-worker serve",
+    ),
 ):
     """Serve a unionml_amd FastAPI app with uvicorn
-    (reference: cli.py:285-320)."""
+    (reference: cli.py:285-320).
+
+    ``--workers N`` uses the framework's own supervisor: N independent
+    single-worker processes sharing the port via SO_REUSEPORT (kernel
+    load balancing), each with TCP_NODELAY set on the listening socket.
+    uvicorn's bundled multiprocess mode is NOT used — its explicit-socket
+    path leaves Nagle enabled, which measures as a uniform ~40 ms
+    delayed-ACK penalty per request on this stack (see
+    profiles/r01_resnet_and_serving.md).
+    """
     if model_path is not None:
         if os.environ.get("UNIONML_MODEL_PATH"):
             typer.echo(
@@ -377,7 +390,50 @@ def serve(
     sys.path.insert(0, os.getcwd())
     import uvicorn
 
-    uvicorn.run(app_spec, host=host, port=port, workers=workers, reload=reload)
+    if reuse_port_worker:
+        import socket as socket_mod
+
+        sock = socket_mod.socket(socket_mod.AF_INET, socket_mod.SOCK_STREAM)
+        sock.setsockopt(socket_mod.SOL_SOCKET, socket_mod.SO_REUSEADDR, 1)
+        sock.setsockopt(socket_mod.SOL_SOCKET, socket_mod.SO_REUSEPORT, 1)
+        # accepted sockets inherit this on Linux; uvicorn's passed-socket
+        # path never sets it itself
+        sock.setsockopt(socket_mod.IPPROTO_TCP, socket_mod.TCP_NODELAY, 1)
+        sock.bind((host, port))
+        sock.listen(2048)
+        config = uvicorn.Config(app_spec, log_level="info")
+        uvicorn.Server(config).run(sockets=[sock])
+        return
+
+    if workers > 1:
+        import signal
+        import subprocess
+
+        children = []
+        cmd = [
+            sys.executable, "-m", "unionml_amd.cli", "serve", app_spec,
+            "--host", host, "--port", str(port), "--reuse-port-worker",
+        ]
+        env = dict(os.environ)
+        for _ in range(workers):
+            children.append(subprocess.Popen(cmd, env=env))
+
+        def shutdown(signum, frame):
+            for c in children:
+                c.terminate()
+
+        signal.signal(signal.SIGTERM, shutdown)
+        signal.signal(signal.SIGINT, shutdown)
+        try:
+            for c in children:
+                c.wait()
+        finally:
+            for c in children:
+                if c.poll() is None:
+                    c.terminate()
+        return
+
+    uvicorn.run(app_spec, host=host, port=port, reload=reload)
 
 
 def main():
