@@ -93,6 +93,9 @@ def main():
         env.pop("UNIONML_MODEL_PATH", None)
         repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
         env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        # stdout MUST not be an undrained pipe: uvicorn logs would fill the
+        # pipe buffer under load and deadlock the server
+        server_log = open(os.path.join(tmp, "server.log"), "wb")
         proc = subprocess.Popen(
             [
                 sys.executable, "-m", "unionml_amd.cli", "serve",
@@ -100,7 +103,7 @@ def main():
                 "--model-path", artifact_path,
                 "--port", str(port), "--workers", str(args.workers),
             ],
-            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, env=env,
+            stdout=server_log, stderr=subprocess.STDOUT, env=env,
         )
     else:
         app = FastAPI()
@@ -113,7 +116,11 @@ def main():
     deadline = time.monotonic() + 120
     while time.monotonic() < deadline:
         if proc is not None and proc.poll() is not None:
-            print(proc.stdout.read().decode()[-3000:], file=sys.stderr)
+            try:
+                with open(server_log.name, "rb") as fh:
+                    print(fh.read().decode()[-3000:], file=sys.stderr)
+            except OSError:
+                pass
             sys.exit(1)
         try:
             if httpx.get(f"{url}/health", timeout=1.0).status_code == 200:

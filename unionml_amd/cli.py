@@ -401,7 +401,7 @@ def serve(
         sock.setsockopt(socket_mod.IPPROTO_TCP, socket_mod.TCP_NODELAY, 1)
         sock.bind((host, port))
         sock.listen(2048)
-        config = uvicorn.Config(app_spec, log_level="info")
+        config = uvicorn.Config(app_spec, log_level="warning", access_log=False)
         uvicorn.Server(config).run(sockets=[sock])
         return
 
