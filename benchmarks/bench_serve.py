@@ -55,6 +55,29 @@ def pct(xs, q):
     return xs[min(len(xs) - 1, int(q * len(xs)))]
 
 
+def _load_proc(q, url, one_row, clients, reqs_per_client):
+    """One load-generator process: `clients` async clients, each posting
+    `reqs_per_client` sequential requests."""
+    import httpx
+
+    async def run():
+        async with httpx.AsyncClient(base_url=url, timeout=60.0) as client:
+
+            async def worker(n):
+                times = []
+                for _ in range(n):
+                    t0 = time.perf_counter()
+                    r = await client.post("/predict", json={"features": one_row})
+                    times.append((time.perf_counter() - t0) * 1000.0)
+                    assert r.status_code == 200
+                return times
+
+            out = await asyncio.gather(*(worker(reqs_per_client) for _ in range(clients)))
+            return [t for ts in out for t in ts]
+
+    q.put(asyncio.run(run()))
+
+
 def main():
     args = parse_args()
     import torch
@@ -144,26 +167,32 @@ def main():
             lat.append((time.perf_counter() - t0) * 1000.0)
             assert r.status_code == 200, r.text
 
-    # --- concurrent load ----------------------------------------------------
-    async def load():
-        async with httpx.AsyncClient(base_url=url, timeout=30.0) as client:
+    # --- concurrent load (multi-process generator: one asyncio client
+    # process saturates at ~650 req/s — far below the server) ---------------
+    import multiprocessing as mp
 
-            async def worker(n):
-                times = []
-                for _ in range(n):
-                    t0 = time.perf_counter()
-                    r = await client.post("/predict", json={"features": one_row})
-                    times.append((time.perf_counter() - t0) * 1000.0)
-                    assert r.status_code == 200
-                return times
+    n_procs = min(8, max(1, args.concurrency // 4))
+    clients_per_proc = max(1, args.concurrency // n_procs)
+    reqs_per_client = max(1, args.requests // (n_procs * clients_per_proc))
 
-            per = max(1, args.requests // args.concurrency)
-            out = await asyncio.gather(*(worker(per) for _ in range(args.concurrency)))
-            return [t for ts in out for t in ts]
-
+    ctx = mp.get_context("spawn")  # no CUDA state in the generator procs
+    queue = ctx.Queue()
+    procs = [
+        ctx.Process(
+            target=_load_proc,
+            args=(queue, url, one_row, clients_per_proc, reqs_per_client),
+        )
+        for _ in range(n_procs)
+    ]
     t0 = time.perf_counter()
-    conc_lat = asyncio.run(load())
+    for p in procs:
+        p.start()
+    conc_lat = []
+    for _ in procs:
+        conc_lat.extend(queue.get())
     conc_wall = time.perf_counter() - t0
+    for p in procs:
+        p.join()
 
     # --- raw in-process graph replay (no HTTP) ------------------------------
     raw_p50 = None
