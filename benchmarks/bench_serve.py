@@ -109,7 +109,10 @@ def main():
         import subprocess
         import tempfile
 
-        tmp = tempfile.mkdtemp(prefix="unionml_serve_bench_")
+        tmp = os.environ.get("UNIONML_BENCH_DIR") or tempfile.mkdtemp(
+            prefix="unionml_serve_bench_"
+        )
+        os.makedirs(tmp, exist_ok=True)
         artifact_path = os.path.join(tmp, "model.pt")
         model.save(artifact_path)
         env = dict(os.environ)
