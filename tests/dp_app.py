@@ -135,3 +135,30 @@ def run_tabular_dp():
         with open(Path(out_dir) / "dp_master.pkl", "rb") as f:
             dp_master = pickle.load(f)
     return single_master, dp_master
+
+
+def build_failing_model():
+    """An app whose trainer raises — for DP failure-surfacing tests."""
+    ds = Dataset(name="fail_ds", targets=["y"], test_size=0.2, random_state=5)
+
+    @ds.reader
+    def reader(n: int = 40) -> pd.DataFrame:
+        import numpy as np
+
+        rng = np.random.RandomState(0)
+        X = rng.rand(n, 3)
+        return pd.DataFrame(
+            {"x1": X[:, 0], "x2": X[:, 1], "x3": X[:, 2], "y": (X.sum(axis=1) > 1.5).astype(int)}
+        )
+
+    m = Model(name="fail_model", init=DPNet, dataset=ds)
+
+    @m.trainer
+    def trainer(net: DPNet, features: pd.DataFrame, target: pd.DataFrame) -> DPNet:
+        raise RuntimeError("injected trainer failure")
+
+    @m.predictor
+    def predictor(net: DPNet, features: pd.DataFrame) -> List[int]:
+        return [0] * len(features)
+
+    return m

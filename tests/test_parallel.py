@@ -162,3 +162,13 @@ def test_reducer_broadcasts_initial_params(tmp_path):
         start_method="spawn",
     )
     assert (tmp_path / "ok_0").exists() and (tmp_path / "ok_1").exists()
+
+
+def test_dp_worker_failure_surfaces_clean_error():
+    """A trainer crashing inside a DP worker must surface as a
+    RuntimeError with the worker traceback, not a hang."""
+    from dp_app import build_failing_model
+
+    model = build_failing_model()
+    with pytest.raises(RuntimeError, match="injected trainer failure"):
+        model.train(dp=2, n=40)

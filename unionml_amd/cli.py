@@ -309,6 +309,25 @@ def list_scheduled_prediction_runs(
         typer.echo(run)
 
 
+@app.command("run-scheduler")
+def run_scheduler(
+    model_spec: str = typer.Argument(...),
+    app_version: Optional[str] = typer.Option(None, "--app-version", "-v"),
+    iterations: Optional[int] = typer.Option(
+        None, "--iterations", "-n", help="bound the loop (default: run forever)"
+    ),
+    poll_s: float = typer.Option(1.0, "--poll-s"),
+):
+    """Run the backend scheduler loop, firing active cron / fixed-rate
+    launch plans (the reference delegates this to Flyte; here the
+    framework owns it — see docs/scheduling.md)."""
+    model = get_model(model_spec)
+    typer.echo("scheduler running (ctrl-c to stop)")
+    model._backend().run_scheduler(
+        model, app_version=app_version, iterations=iterations, poll_s=poll_s
+    )
+
+
 # ----------------------------------------------------------------------
 # fetch
 # ----------------------------------------------------------------------

@@ -140,13 +140,18 @@ def train_data_parallel(
     port = _free_port()
 
     with tempfile.TemporaryDirectory(prefix="unionml_amd_dp_") as out_dir:
-        mp.start_processes(
-            _dp_worker,
-            args=(dp, port, model_blob, kwargs_blob, out_dir),
-            nprocs=dp,
-            join=True,
-            start_method="spawn",
-        )
+        try:
+            mp.start_processes(
+                _dp_worker,
+                args=(dp, port, model_blob, kwargs_blob, out_dir),
+                nprocs=dp,
+                join=True,
+                start_method="spawn",
+            )
+        except Exception as exc:
+            errors = sorted(Path(out_dir).glob("error_rank*.txt"))
+            detail = errors[0].read_text() if errors else str(exc)
+            raise RuntimeError(f"dp worker failed:\n{detail}") from exc
         errors = sorted(Path(out_dir).glob("error_rank*.txt"))
         if errors:
             raise RuntimeError(f"dp worker failed:\n{errors[0].read_text()}")
