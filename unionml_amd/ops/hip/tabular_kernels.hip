@@ -325,6 +325,8 @@ __device__ __forceinline__ void chunk_fwd_bwd(
     s += __shfl_xor(s, 32, 64);
     const float rs = fast_rcp(s);
     const float logs = __logf(s);
+    float db2_acc[4];
+    float loss_acc = 0.f;
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int c = lg * 4 + r;
@@ -332,11 +334,27 @@ __device__ __forceinline__ void chunk_fwd_bwd(
       const u16 dlb = f2bf(dl);
       L.DLs[row][c] = dlb;
       L.DLT[c][row] = dlb;
-      if (valid && c == label) {
-        atomicAdd(L.loss, -(logit[r] - m - logs) * invBtot);
-      }
-      atomicAdd(&L.db2[c], dl);
+      db2_acc[r] = dl;
+      if (valid && c == label) loss_acc = -(logit[r] - m - logs) * invBtot;
     }
+    // reduce db2/loss over the 16 row-lanes (lr bits) first, then ONE
+    // LDS atomic per (wave, class) — the per-element atomic version
+    // serialized ~2048 adds onto 16 LDS addresses per workgroup
+    #pragma unroll
+    for (int bit = 1; bit < 16; bit <<= 1) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) db2_acc[r] += __shfl_xor(db2_acc[r], bit, 64);
+      loss_acc += __shfl_xor(loss_acc, bit, 64);
+    }
+    if (lr == 0) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) atomicAdd(&L.db2[lg * 4 + r], db2_acc[r]);
+    }
+    // loss_acc now holds the 16-row sum within this lg quarter; fold
+    // the four lg groups (lane bits 4,5) too, then one atomic per wave
+    loss_acc += __shfl_xor(loss_acc, 16, 64);
+    loss_acc += __shfl_xor(loss_acc, 32, 64);
+    if (l == 0) atomicAdd(L.loss, loss_acc);
   }
   __syncthreads();   // DLs/DLT complete
 
