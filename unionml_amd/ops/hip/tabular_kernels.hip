@@ -290,7 +290,11 @@ __device__ __forceinline__ void chunk_fwd_bwd(
       L.HT[h][row] = hb;
     }
   }
-  __syncthreads();   // Hs/HT complete
+  // NO __syncthreads here: fwd2 reads only THIS wave's rows of Hs
+  // (row = wrow+lr), written by lanes of the same wave — a wave-level
+  // LDS drain is sufficient. (dW2's cross-wave HT reads are covered by
+  // the barrier after dH below.)
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
   // ---- fwd2: L^T = W2T @ B(Hs);  D: c = lg*4+r, row = wrow+lr --------------
   {
@@ -356,7 +360,10 @@ __device__ __forceinline__ void chunk_fwd_bwd(
     loss_acc += __shfl_xor(loss_acc, 32, 64);
     if (l == 0) atomicAdd(L.loss, loss_acc);
   }
-  __syncthreads();   // DLs/DLT complete
+  // NO __syncthreads here either: dH reads only this wave's rows of
+  // DLs and HT. Cross-wave consumers (dW1/dW2 over DLT/HT) wait at the
+  // barrier after dH.
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
   // ---- dH^T = W2s @ B(DLs);  D: h = mt*16+lg*4+r, row = wrow+lr ------------
   #pragma unroll
