@@ -168,3 +168,23 @@ def test_stage_to_device_cpu(frame_dataset):
     staged = frame_dataset.stage_to_device(data["train"], device="cpu")
     assert torch.is_tensor(staged[0])
     assert staged[0].shape == (8, 3)
+
+
+def test_dataset_from_task():
+    """Dataset.from_task wraps an existing Task/callable as the reader
+    (reference: Dataset._from_flytekit_task, dataset.py:426-440)."""
+    import pandas as pd
+
+    from unionml_amd.dataset import Dataset
+    from unionml_amd.task import Task
+
+    def produce(n: int = 10) -> pd.DataFrame:
+        return pd.DataFrame({"a": range(n), "y": [i % 2 for i in range(n)]})
+
+    task = Task(produce, "produce")
+    ds = Dataset.from_task(task, targets=["y"], test_size=0.2, random_state=0)
+    assert ds.name == "produce"
+    data = ds.get_data(ds.dataset_task()(n=20))
+    (Xtr, ytr), (Xte, yte) = data["train"], data["test"]
+    assert len(Xtr) + len(Xte) == 20
+    assert list(Xtr.columns) == ["a"]

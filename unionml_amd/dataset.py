@@ -285,6 +285,23 @@ class Dataset(TrackedInstance):
     # ------------------------------------------------------------------
 
     @classmethod
+    def from_task(cls, task, name: Optional[str] = None, **dataset_kwargs) -> "Dataset":
+        """Build a Dataset whose reader is an existing :class:`Task` (or
+        any annotated callable) — the analog of the reference's
+        ``Dataset._from_flytekit_task`` (dataset.py:426-440), which the
+        SQL constructors below are built on."""
+        ds = cls(name=name or getattr(task, "name", "dataset"), **dataset_kwargs)
+        fn = task.fn if hasattr(task, "fn") else task
+
+        def task_reader(**kwargs):
+            return task(**kwargs)
+
+        task_reader.__annotations__ = dict(getattr(fn, "__annotations__", {}))
+        task_reader.__signature__ = signature(fn)
+        ds.reader(task_reader)
+        return ds
+
+    @classmethod
     def from_sqlite_task(
         cls,
         name: str,
