@@ -33,7 +33,7 @@ def parse_args():
     p.add_argument("--image-size", type=int, default=224)
     p.add_argument("--classes", type=int, default=1000)
     p.add_argument("--lr", type=float, default=1e-3)
-    p.add_argument("--bucket-mb", type=float, default=64.0)
+    p.add_argument("--bucket-mb", type=float, default=16.0, help="gradient bucket size; ResNet-18 grads are ~47MB fp32, so 16MB gives ~3 buckets overlapping backward")
     p.add_argument("--minibatches", type=int, default=4)
     return p.parse_args()
 
