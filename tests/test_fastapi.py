@@ -284,3 +284,28 @@ def test_serve_subprocess_http(tmp_path):
             server.wait(timeout=10)
         except subprocess.TimeoutExpired:
             server.kill()
+
+
+def test_metrics_endpoint():
+    """GET /metrics exposes Prometheus counters + latency histogram
+    (net-new observability vs the reference's /health-only surface)."""
+    model = build_sklearn_app()
+    model.train()
+    app = FastAPI()
+    model.serve(app)
+    feats = [{"x1": 0.5, "x2": 0.1, "x3": 0.9}, {"x1": 0.2, "x2": 0.8, "x3": 0.4}]
+    with TestClient(app) as client:
+        for _ in range(3):
+            assert client.post("/predict", json={"features": feats}).status_code == 200
+        resp = client.get("/metrics")
+        assert resp.status_code == 200
+        body = resp.text
+        assert "unionml_predict_requests_total" in body
+        assert "unionml_predict_rows_total" in body
+        assert "unionml_predict_latency_seconds_bucket" in body
+        import re
+
+        m = re.search(r'unionml_predict_requests_total{[^}]*} (\d+\.\d+)', body)
+        assert m and float(m.group(1)) == 3.0, body[:500]
+        m = re.search(r'unionml_predict_rows_total{[^}]*} (\d+\.\d+)', body)
+        assert m and float(m.group(1)) == 6.0

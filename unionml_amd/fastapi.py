@@ -7,6 +7,7 @@ bucketed hipGraph captures (unionml_amd/serving/batcher.py).
 """
 
 import os
+import time
 from typing import Any, Dict, List, Optional, Union
 
 from unionml_amd._logging import logger
@@ -24,10 +25,18 @@ def serving_app(
     batch: bool = False,
     max_batch_size: int = 64,
     max_delay_ms: float = 0.0,
+    metrics: bool = True,
 ):
     from fastapi import Body, HTTPException
 
-    state = {"batcher": None}
+    state = {"batcher": None, "metrics": None}
+    if metrics:
+        try:
+            from unionml_amd.serving.metrics import ServingMetrics
+
+            state["metrics"] = ServingMetrics(model.name)
+        except ImportError:  # prometheus_client not installed
+            pass
 
     @app.on_event("startup")
     async def load_model():
@@ -73,21 +82,41 @@ def serving_app(
             raise HTTPException(
                 status_code=400, detail="provide one of 'inputs' or 'features'"
             )
+        mx = state["metrics"]
+        t0 = time.perf_counter() if mx else 0.0
         try:
             if features is not None:
+                n_rows = len(features) if hasattr(features, "__len__") else 1
                 if state["batcher"] is not None:
-                    return await state["batcher"].submit(features)
-                features = model._dataset.get_features(features)
-                wf = model.predict_from_features_workflow()
-                return _jsonable(
-                    wf(model_object=model.artifact.model_object, features=features)
-                )
-            return _jsonable(model.predict(**(inputs or {})))
+                    out = await state["batcher"].submit(features)
+                else:
+                    features = model._dataset.get_features(features)
+                    wf = model.predict_from_features_workflow()
+                    out = _jsonable(
+                        wf(model_object=model.artifact.model_object, features=features)
+                    )
+            else:
+                out = _jsonable(model.predict(**(inputs or {})))
+                n_rows = len(out) if hasattr(out, "__len__") else 1
+            if mx:
+                mx.observe(n_rows, time.perf_counter() - t0)
+            return out
         except HTTPException:
             raise
         except Exception as exc:
+            if mx:
+                mx.error()
             logger.exception("prediction failed")
             raise HTTPException(status_code=500, detail=str(exc))
+
+    if state["metrics"] is not None:
+
+        @app.get("/metrics")
+        async def metrics_route():
+            from fastapi import Response
+
+            payload, content_type = state["metrics"].render()
+            return Response(content=payload, media_type=content_type)
 
     @app.get("/health")
     async def health():
