@@ -165,7 +165,9 @@ standardize_apply_kernel(const float* __restrict__ X, long long n_elems, int D,
 #define C_DB1  ALIGN16(C_W2T + CPAD * WS * 2)
 #define C_DB2  ALIGN16(C_DB1 + HID * 4)
 #define C_LOSS ALIGN16(C_DB2 + CPAD * 4)
-#define C_IMG_TOTAL ALIGN16(C_LOSS + 16)
+#define C_B1S  ALIGN16(C_LOSS + 16)
+#define C_B2S  ALIGN16(C_B1S + HID * 4)
+#define C_IMG_TOTAL ALIGN16(C_B2S + CPAD * 4)
 // steps kernel appends optimizer state after the images:
 #define C_MASTER C_IMG_TOTAL
 #define C_M    ALIGN16(C_MASTER + NPARAM * 4)
@@ -186,6 +188,8 @@ struct Lds {
   float* db1;
   float* db2;
   float* loss;
+  float* b1s;   // bias prefetch (fwd chain must not stall on cold HBM)
+  float* b2s;
 };
 
 __device__ __forceinline__ Lds carve(char* smem) {
@@ -203,7 +207,20 @@ __device__ __forceinline__ Lds carve(char* smem) {
   L.db1 = (float*)(smem + C_DB1);
   L.db2 = (float*)(smem + C_DB2);
   L.loss = (float*)(smem + C_LOSS);
+  L.b1s = (float*)(smem + C_B1S);
+  L.b2s = (float*)(smem + C_B2S);
   return L;
+}
+
+// prefetch biases into LDS alongside the other prologue loads — the
+// previous launch's agent-scope acquire left L2 cold, so a lazy mid-GEMM
+// b1/b2 read would serialize a full HBM round trip into the fwd chain
+__device__ __forceinline__ void load_biases(const Lds& L,
+                                            const float* __restrict__ b1,
+                                            const float* __restrict__ b2) {
+  const int tid = threadIdx.x;
+  if (tid < HID) L.b1s[tid] = b1[tid];
+  else if (tid < HID + CPAD) L.b2s[tid - HID] = b2[tid - HID];
 }
 
 // fill weight images (W1T/W2s/W2T incl. zero K-pads) from bf16 weight arrays
@@ -439,12 +456,13 @@ mlp_step_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y, int B,
   zero_dl_pad(L);
   load_weight_images(L, W1bf, W2bf);
   load_x_chunk(Xbf, L, row0, B);
+  load_biases(L, master + OFF_B1, master + OFF_B2);
   __syncthreads();
 
   ChunkAcc acc;
   acc.dW1 = (f32x4){0.f, 0.f, 0.f, 0.f};
   acc.dW2 = (f32x4){0.f, 0.f, 0.f, 0.f};
-  chunk_fwd_bwd(L, master + OFF_B1, master + OFF_B2, y, row0, B, invBtot, acc);
+  chunk_fwd_bwd(L, L.b1s, L.b2s, y, row0, B, invBtot, acc);
 
   {
     const int ht = wave & 1, it = wave >> 1;
@@ -523,13 +541,14 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   load_weight_images(L, W1bf, W2bf);
   load_x_chunk(Xbf, L, row0, B);
 #endif
+  load_biases(L, master + OFF_B1, master + OFF_B2);
   __syncthreads();
 
   ChunkAcc acc;
   acc.dW1 = (f32x4){0.f, 0.f, 0.f, 0.f};
   acc.dW2 = (f32x4){0.f, 0.f, 0.f, 0.f};
 #ifndef PROBE_SKIP_FWDBWD
-  chunk_fwd_bwd(L, master + OFF_B1, master + OFF_B2, y, row0, B, invBtot, acc);
+  chunk_fwd_bwd(L, L.b1s, L.b2s, y, row0, B, invBtot, acc);
 #endif
 
   // ---- write this WG's complete partial slab (plain stores, no atomics) ----
