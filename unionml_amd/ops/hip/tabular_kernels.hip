@@ -88,6 +88,55 @@ __device__ __forceinline__ float fast_rcp(float x) {
 // standardize
 // ---------------------------------------------------------------------------
 
+// fast path (256 % D == 0): coalesced row-major accumulation — thread t
+// owns column t%D and row-group t/D, so each iteration reads 256
+// consecutive floats; per-column partials reduce in LDS and land in a
+// double scratch via atomics; a tiny finalize kernel produces
+// mean/invstd. (The column-per-workgroup variant below read the matrix
+// with stride D — 181 µs for digits vs ~8 µs here.)
+extern "C" __global__ void __launch_bounds__(256)
+standardize_fit_fast_kernel(const float* __restrict__ X, long long N, int D,
+                            double* __restrict__ scratch /* [2*D] zeroed */) {
+  const int tid = threadIdx.x;
+  const int col = tid % D;
+  const int rg = tid / D;
+  const int rpi = 256 / D;                 // rows per iteration per WG
+  double s = 0.0, s2 = 0.0;
+  for (long long r = (long long)blockIdx.x * rpi + rg; r < N;
+       r += (long long)gridDim.x * rpi) {
+    const double v = (double)X[r * D + col];
+    s += v;
+    s2 += v * v;
+  }
+  __shared__ double ls[256], ls2[256];
+  ls[tid] = s;
+  ls2[tid] = s2;
+  __syncthreads();
+  for (int off = 128; off >= D; off >>= 1) {
+    if (tid < off && tid + off < 256) {
+      ls[tid] += ls[tid + off];
+      ls2[tid] += ls2[tid + off];
+    }
+    __syncthreads();
+  }
+  if (tid < D) {
+    atomicAdd(&scratch[col], ls[tid]);
+    atomicAdd(&scratch[D + col], ls2[tid]);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+standardize_fit_finalize_kernel(const double* __restrict__ scratch, long long N,
+                                int D, float* __restrict__ mean,
+                                float* __restrict__ invstd, float eps) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= D) return;
+  const double m = scratch[col] / (double)N;
+  const double var = scratch[D + col] / (double)N - m * m;
+  mean[col] = (float)m;
+  invstd[col] = (float)(1.0 / sqrt(var > 0.0 ? var + (double)eps : (double)eps));
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 standardize_fit_kernel(const float* __restrict__ X, long long N, int D,
                        float* __restrict__ mean, float* __restrict__ invstd,
@@ -950,7 +999,20 @@ static int set_lds(const void* kernel, int bytes) {
 extern "C" {
 
 void launch_standardize_fit(const float* X, long long N, int D, float* mean,
-                            float* invstd, float eps, hipStream_t stream) {
+                            float* invstd, float eps, double* scratch,
+                            hipStream_t stream) {
+  if (scratch != nullptr && D <= 256 && 256 % D == 0) {
+    const int rpi = 256 / D;
+    int blocks = (int)((N + rpi - 1) / rpi);
+    if (blocks > 128) blocks = 128;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(standardize_fit_fast_kernel, dim3(blocks), dim3(256), 0,
+                       stream, X, N, D, scratch);
+    hipLaunchKernelGGL(standardize_fit_finalize_kernel,
+                       dim3((D + 255) / 256), dim3(256), 0, stream, scratch, N,
+                       D, mean, invstd, eps);
+    return;
+  }
   hipLaunchKernelGGL(standardize_fit_kernel, dim3(D), dim3(256), 0, stream,
                      X, N, D, mean, invstd, eps);
 }

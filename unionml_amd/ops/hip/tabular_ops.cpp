@@ -7,7 +7,7 @@
 
 extern "C" {
 void launch_standardize_fit(const float*, long long, int, float*, float*, float,
-                            hipStream_t);
+                            double*, hipStream_t);
 void launch_standardize_apply(const float*, long long, int, const float*,
                               const float*, unsigned short*, hipStream_t);
 void launch_mlp_step(const unsigned short*, const int*, int, const unsigned short*,
@@ -57,8 +57,15 @@ void standardize_fit(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd,
   check(invstd, torch::kFloat32, "invstd");
   const int D = (int)X.size(1);
   TORCH_CHECK(mean.numel() == D && invstd.numel() == D, "mean/invstd size mismatch");
+  double* scratch_ptr = nullptr;
+  torch::Tensor scratch;
+  if (D <= 256 && 256 % D == 0) {
+    scratch = torch::zeros({2 * D}, mean.options().dtype(torch::kFloat64));
+    scratch_ptr = scratch.data_ptr<double>();
+  }
   launch_standardize_fit(X.data_ptr<float>(), X.size(0), D, mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), (float)eps, current_stream());
+                         invstd.data_ptr<float>(), (float)eps, scratch_ptr,
+                         current_stream());
 }
 
 void standardize_apply(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd,
