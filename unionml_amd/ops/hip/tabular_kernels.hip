@@ -1004,7 +1004,7 @@ void launch_standardize_fit(const float* X, long long N, int D, float* mean,
   if (scratch != nullptr && D <= 256 && 256 % D == 0) {
     const int rpi = 256 / D;
     int blocks = (int)((N + rpi - 1) / rpi);
-    if (blocks > 128) blocks = 128;
+    if (blocks > 1024) blocks = 1024;  // >256 WGs fills all 8 XCDs; atomics stay cheap
     if (blocks < 1) blocks = 1;
     hipLaunchKernelGGL(standardize_fit_fast_kernel, dim3(blocks), dim3(256), 0,
                        stream, X, N, D, scratch);
