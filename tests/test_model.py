@@ -181,3 +181,72 @@ def test_prediction_callbacks_swallow_errors():
     preds = model.predict(features=[{"x1": 0, "x2": 0, "x3": 0}])
     assert len(preds) == 1
     assert calls == [1]  # good callback ran, bad one was swallowed
+
+
+def test_hyperparameter_type_resolution_branches():
+    """The 4-branch hyperparameter-type logic (reference model.py:168-204):
+    explicit > dataclass annotation on the init arg > synthesized from
+    init keyword annotations > dict."""
+    import dataclasses
+
+    from model_fixtures import make_dataset
+    from unionml_amd import Model
+
+    @dataclasses.dataclass
+    class HP:
+        lr: float = 0.1
+        depth: int = 2
+
+    # 1: explicit
+    m1 = Model(name="m1", hyperparameter_type=HP, dataset=make_dataset())
+    assert m1.hyperparameter_type is HP
+
+    # 2: dataclass annotation on the init arg
+    m2 = Model(name="m2", dataset=make_dataset())
+
+    @m2.init
+    def init2(hyperparameters: HP) -> object:
+        return object()
+
+    assert m2.hyperparameter_type is HP
+
+    # 4: nothing declared -> dict
+    m4 = Model(name="m4", dataset=make_dataset())
+    assert m4.hyperparameter_type is dict
+
+    # coercion: dict -> dataclass instance for branch 1/2
+    hp = m1._coerce_hyperparameters({"lr": 0.5, "depth": 3})
+    assert isinstance(hp, HP) and hp.lr == 0.5 and hp.depth == 3
+
+
+def test_init_function_keyword_only_style():
+    """Branch 3: an init FUNCTION with keyword-only annotated params
+    synthesizes a hyperparameter dataclass and is called with the
+    hyperparameters spread as kwargs."""
+    import dataclasses
+
+    from sklearn.linear_model import LogisticRegression
+
+    from model_fixtures import make_dataset
+    from unionml_amd import Model
+
+    m = Model(name="kw_init", dataset=make_dataset())
+
+    @m.init
+    def init(*, C: float = 1.0, max_iter: int = 200) -> LogisticRegression:
+        return LogisticRegression(C=C, max_iter=max_iter)
+
+    hp_t = m.hyperparameter_type
+    assert dataclasses.is_dataclass(hp_t)
+    assert set(f.name for f in dataclasses.fields(hp_t)) == {"C", "max_iter"}
+
+    @m.trainer
+    def trainer(est: LogisticRegression, features, target) -> LogisticRegression:
+        return est.fit(features, target.squeeze())
+
+    @m.predictor
+    def predictor(est: LogisticRegression, features) -> list:
+        return [float(x) for x in est.predict(features)]
+
+    model_obj, _ = m.train(hyperparameters={"C": 0.5, "max_iter": 300})
+    assert model_obj.C == 0.5 and model_obj.max_iter == 300

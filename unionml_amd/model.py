@@ -121,6 +121,19 @@ class Model(TrackedInstance):
             hp_param = params.get("hyperparameters")
             if hp_param is not None and dataclasses.is_dataclass(hp_param.annotation):
                 return hp_param.annotation
+            # synthesize from an init FUNCTION's keyword-only annotations
+            # (reference branch 3, model.py:168-204)
+            kw_fields = []
+            for pname, p in params.items():
+                if p.kind is not p.KEYWORD_ONLY or pname == "hyperparameters":
+                    continue
+                ann = p.annotation if p.annotation is not inspect.Parameter.empty else Any
+                if p.default is not inspect.Parameter.empty:
+                    kw_fields.append((pname, ann, field(default=p.default)))
+                else:
+                    kw_fields.append((pname, ann))
+            if kw_fields:
+                return make_dataclass(f"{type(self).__name__}Hyperparameters", kw_fields)
 
         if inspect.isclass(init):
             fields = []
@@ -272,18 +285,23 @@ class Model(TrackedInstance):
         return {"hyperparameters": hyperparameters}
 
     def _call_init(self, hyperparameters) -> Any:
-        """Create a fresh model object (reference: model.py:1425-1430)."""
+        """Create a fresh model object (reference: model.py:1425-1430).
+
+        Function inits are called ``init(hyperparameters=dict)`` when
+        they declare a ``hyperparameters`` parameter, else with the
+        hyperparameters spread as keyword arguments (the synthesized-
+        dataclass style, reference branch 3)."""
         hp_kwargs = self._hyperparameters_as_kwargs(hyperparameters)
-        if self._init_fn is not None:
-            return self._init_fn(hyperparameters=hp_kwargs)
-        init = self._init_cls_or_fn
+        init = self._init_fn or self._init_cls_or_fn
         if init is None:
             raise ValueError(
                 f"model '{self.name}' has no init: pass init= to Model(...) or use @model.init"
             )
         if inspect.isclass(init):
             return init(**hp_kwargs)
-        return init(hyperparameters=hp_kwargs)
+        if "hyperparameters" in signature(init).parameters:
+            return init(hyperparameters=hp_kwargs)
+        return init(**hp_kwargs)
 
     # ------------------------------------------------------------------
     # compiled tasks
