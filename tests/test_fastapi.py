@@ -309,3 +309,49 @@ def test_metrics_endpoint():
         assert m and float(m.group(1)) == 3.0, body[:500]
         m = re.search(r'unionml_predict_rows_total{[^}]*} (\d+\.\d+)', body)
         assert m and float(m.group(1)) == 6.0
+
+
+def test_reload_endpoint(tmp_path, monkeypatch):
+    """POST /reload hot-swaps the artifact from $UNIONML_MODEL_PATH."""
+    model = build_sklearn_app()
+    model.train()
+    path_a = tmp_path / "a.joblib"
+    model.save(path_a)
+    coef_a = model.artifact.model_object.coef_.copy()
+
+    # a second, differently-trained artifact
+    model2 = build_sklearn_app()
+    model2.train(hyperparameters={"C": 0.01})
+    path_b = tmp_path / "b.joblib"
+    model2.save(path_b)
+
+    app = FastAPI()
+    model.serve(app)
+    monkeypatch.setenv("UNIONML_MODEL_PATH", str(path_a))
+    with TestClient(app) as client:
+        assert client.get("/health").status_code == 200
+        import numpy as np
+
+        assert np.allclose(model.artifact.model_object.coef_, coef_a)
+
+        monkeypatch.setenv("UNIONML_MODEL_PATH", str(path_b))
+        resp = client.post("/reload")
+        assert resp.status_code == 200, resp.text
+        assert not np.allclose(model.artifact.model_object.coef_, coef_a)
+        # serving still works on the new artifact
+        feats = [{"x1": 0.5, "x2": 0.1, "x3": 0.9}]
+        assert client.post("/predict", json={"features": feats}).status_code == 200
+
+    monkeypatch.delenv("UNIONML_MODEL_PATH", raising=False)
+
+
+def test_reload_without_source_400():
+    model = build_sklearn_app()
+    model.train()
+    app = FastAPI()
+    model.serve(app)
+    import os
+
+    os.environ.pop("UNIONML_MODEL_PATH", None)
+    with TestClient(app) as client:
+        assert client.post("/reload").status_code == 400

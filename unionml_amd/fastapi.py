@@ -124,6 +124,42 @@ def serving_app(
             raise HTTPException(status_code=500, detail="model artifact not loaded")
         return {"status": "ok"}
 
+    @app.post("/reload")
+    async def reload_model():
+        """Hot-reload the artifact (after e.g. `unionml-amd fetch-model`)
+        without restarting the server: re-read $UNIONML_MODEL_PATH (or
+        the backend's latest), swap the artifact, and rebuild the
+        batcher's hipGraphs against the new weights."""
+        model_path = os.environ.get("UNIONML_MODEL_PATH")
+        try:
+            if model_path:
+                model.artifact = ModelArtifact(model.load(model_path))
+            elif remote:
+                model.artifact = model._backend().fetch_model_artifact(
+                    model, app_version=app_version, model_version=model_version
+                )
+            else:
+                raise HTTPException(
+                    status_code=400,
+                    detail="no reload source: set UNIONML_MODEL_PATH or serve with remote=True",
+                )
+        except HTTPException:
+            raise
+        except Exception as exc:
+            logger.exception("model reload failed")
+            raise HTTPException(status_code=500, detail=str(exc))
+        if batch:
+            from unionml_amd.serving.batcher import DynamicBatcher
+
+            old = state["batcher"]
+            state["batcher"] = DynamicBatcher(
+                model, max_batch_size=max_batch_size, max_delay_ms=max_delay_ms
+            )
+            state["batcher"].start()
+            if old is not None:
+                old.stop()
+        return {"status": "reloaded", "source": model_path or "backend"}
+
     return app
 
 
