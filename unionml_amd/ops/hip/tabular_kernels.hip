@@ -628,6 +628,13 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   // scheme is robust to any grid size, graph replay, and engine mixing;
   // no counter, no reset. ----------------------------------------------------
   const unsigned epoch = lossu[2] + 1u;
+  // Adam-state prefetch scratch (loads issued during the sweep below);
+  // the chunk arrays are dead from here on — reuse Xs
+  float* pre = (float*)L.Xs;
+  const int span_pre = (NPARAM + 1 + n_wg - 1) / n_wg;
+  const int lo_pre = blockIdx.x * span_pre;
+  const bool use_pre =
+      (grads_out == nullptr) && (3 * span_pre * 4 <= ROWS * XS * 2);
 #ifndef PROBE_SKIP_HANDSHAKE
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");   // EVERY storing wave drains
   __syncthreads();
@@ -638,6 +645,22 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
                        __HIP_MEMORY_SCOPE_AGENT);
   }
   __syncthreads();                                   // tag stored before polls
+
+  // prefetch this WG's Adam-state stripe (master/m/v) into LDS WHILE the
+  // sweep waits — these are only ever written by this same WG's stripe
+  // in the reduce phase, so reading them before the acquire is safe, and
+  // it pulls an otherwise-cold post-invalidate HBM round trip out of the
+  // reduce chain.
+  if (use_pre) {
+    const int hi_pre = min(lo_pre + span_pre, NPARAM);
+    for (int i = lo_pre + tid; i < hi_pre; i += BLOCK) {
+      const int j = i - lo_pre;
+      pre[j] = master[i];
+      pre[span_pre + j] = m[i];
+      pre[2 * span_pre + j] = v[i];
+    }
+  }
+
   if (wave == 0) {
     // relaxed PARALLEL sweep: lane j polls tags j, j+64, ... (G16: never
     // poll with an acquire). One vector gather per poll round — one
@@ -707,11 +730,15 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
       *loss_out = g;
       continue;
     }
-    const float mi = beta1 * m[i] + (1.f - beta1) * g;
-    const float vi = beta2 * v[i] + (1.f - beta2) * g * g;
+    const int j = i - lo;
+    const float p_old = use_pre ? pre[j] : master[i];
+    const float m_old = use_pre ? pre[span_pre + j] : m[i];
+    const float v_old = use_pre ? pre[2 * span_pre + j] : v[i];
+    const float mi = beta1 * m_old + (1.f - beta1) * g;
+    const float vi = beta2 * v_old + (1.f - beta2) * g * g;
     m[i] = mi;
     v[i] = vi;
-    const float p = master[i] - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
+    const float p = p_old - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
     master[i] = p;
     bfmirror[i] = f2bf(p);
   }
