@@ -89,6 +89,14 @@ def test_quickstart_journey_on_gpu(tmp_path, monkeypatch):
         assert r.status_code == 200, r.text
         preds = r.json()
         assert len(preds) == 1 and 0 <= preds[0] <= 9
+
+        # /metrics shows the request; /reload hot-swaps the artifact
+        m = httpx.get(f"{url}/metrics", timeout=10.0)
+        assert m.status_code == 200 and "unionml_predict_requests_total" in m.text
+        r = httpx.post(f"{url}/reload", timeout=60.0)
+        assert r.status_code == 200, r.text
+        r = httpx.post(f"{url}/predict", json={"features": feats}, timeout=15.0)
+        assert r.status_code == 200, r.text
     finally:
         server.terminate()
         try:
