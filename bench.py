@@ -103,7 +103,15 @@ def main():
             # fire mid-capture); harmless for a short bench
             os.environ.setdefault("NCCL_ASYNC_ERROR_HANDLING", "0")
             os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "0")
-        dist.init_process_group("nccl" if use_gpu else "gloo")
+        try:
+            # bound-device init: eager communicator creation, cleaner
+            # rendezvous on one-process-per-GPU launches
+            dist.init_process_group(
+                "nccl" if use_gpu else "gloo",
+                device_id=torch.device("cuda", local_rank) if use_gpu else None,
+            )
+        except TypeError:  # older torch without device_id
+            dist.init_process_group("nccl" if use_gpu else "gloo")
 
     from unionml_amd.ops.tabular import ADAM_BETA1, ADAM_BETA2, ADAM_EPS, TabularMLP
     from unionml_amd.ops.reference import NPARAM
