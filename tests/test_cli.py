@@ -205,3 +205,60 @@ def test_bentoml_service_gated_import():
 class _DummyModel:
     name = "dummy"
     artifact = None
+
+
+APP_SCHED = APP_SOURCE.replace(
+    'model.remote(backend_path=r"{backend_path}")',
+    'model.remote(backend_path=r"{backend_path}")\n'
+    "from datetime import timedelta\n"
+    'model.schedule_training("retrain", fixed_rate=timedelta(minutes=5), inputs={{"n": 30}})\n'
+    'model.schedule_prediction("batch_pred", expression="0 * * * *", inputs={{"n": 10}},'
+    " activate_on_deploy=False)",
+)
+
+
+@pytest.fixture
+def sched_cli_app(tmp_path, monkeypatch):
+    backend = tmp_path / "backend"
+    (tmp_path / "sched_cli_app.py").write_text(APP_SCHED.format(backend_path=backend))
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.syspath_prepend(str(tmp_path))
+    yield tmp_path
+    sys.modules.pop("sched_cli_app", None)
+
+
+def test_schedule_lifecycle_via_cli(sched_cli_app):
+    """deploy -> activate/deactivate-schedules -> scheduled-run listings
+    (the reference CLI's schedule surface, with its deactivate/list bugs
+    fixed — SURVEY.md §8 quirks)."""
+    import json as _json
+
+    r = runner.invoke(app, ["deploy", "sched_cli_app:model", "--allow-uncommitted"])
+    assert r.exit_code == 0, r.output
+
+    import sched_cli_app as appmod
+
+    backend = appmod.model._backend()
+    manifest = backend._manifest(None)
+    states = {lp["name"]: lp["active"] for lp in manifest["launchplans"]}
+    assert states == {"retrain": True, "batch_pred": False}
+
+    r = runner.invoke(app, ["activate-schedules", "sched_cli_app:model", "batch_pred"])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(app, ["deactivate-schedules", "sched_cli_app:model", "retrain"])
+    assert r.exit_code == 0, r.output
+    manifest = backend._manifest(None)
+    states = {lp["name"]: lp["active"] for lp in manifest["launchplans"]}
+    assert states == {"retrain": False, "batch_pred": True}
+
+    # deactivate all
+    r = runner.invoke(app, ["deactivate-schedules", "sched_cli_app:model"])
+    assert r.exit_code == 0, r.output
+    manifest = backend._manifest(None)
+    assert all(not lp["active"] for lp in manifest["launchplans"])
+
+    # run listings stay empty but exit cleanly
+    for cmd in ("list-scheduled-training-runs", "list-scheduled-prediction-runs"):
+        name = "retrain" if "training" in cmd else "batch_pred"
+        r = runner.invoke(app, [cmd, "sched_cli_app:model", name])
+        assert r.exit_code == 0, r.output
