@@ -27,7 +27,7 @@ def serving_app(
     max_delay_ms: float = 0.0,
     metrics: bool = True,
 ):
-    from fastapi import Body, HTTPException
+    from fastapi import HTTPException, Request
 
     state = {"batcher": None, "metrics": None}
     if metrics:
@@ -72,10 +72,19 @@ def serving_app(
         }
 
     @app.post("/predict")
-    async def predict(
-        inputs: Optional[Dict[str, Any]] = Body(default=None),
-        features: Optional[Any] = Body(default=None),
-    ):
+    async def predict(request: Request):
+        # parse the body once ourselves — the dual-Body(...) signature
+        # costs a pydantic validation pass per request on the hot path
+        try:
+            body = await request.json()
+        except Exception:
+            raise HTTPException(status_code=400, detail="body must be JSON")
+        if not isinstance(body, dict):
+            raise HTTPException(
+                status_code=400, detail="body must be a JSON object"
+            )
+        inputs = body.get("inputs")
+        features = body.get("features")
         if model.artifact is None:
             raise HTTPException(status_code=500, detail="model artifact not loaded")
         if inputs is None and features is None:
