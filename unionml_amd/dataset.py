@@ -17,15 +17,14 @@ MI355X-native additions:
 
 import json
 import sqlite3
-from inspect import Parameter, signature
+from inspect import signature
 from pathlib import Path
-from typing import Any, Callable, Dict, List, NamedTuple, Optional, Tuple, Type
+from typing import Any, Callable, Dict, List, Optional, Tuple, Type
 
 import numpy as np
 import pandas as pd
 
 from unionml_amd import type_guards
-from unionml_amd._logging import logger
 from unionml_amd.defaults import DEFAULT_RESOURCES, Resources
 from unionml_amd.task import Task, inner_task
 from unionml_amd.tracker import TrackedInstance

@@ -8,11 +8,10 @@ bucketed hipGraph captures (unionml_amd/serving/batcher.py).
 
 import os
 import time
-from typing import Any, Dict, List, Optional, Union
+from typing import Optional
 
 from unionml_amd._logging import logger
 from unionml_amd.artifact import ModelArtifact
-from unionml_amd.exceptions import ModelArtifactNotFound
 
 
 def serving_app(

@@ -22,7 +22,7 @@ import os
 from dataclasses import dataclass, field, make_dataclass
 from inspect import signature
 from pathlib import Path
-from typing import Any, Callable, Dict, IO, List, NamedTuple, Optional, Tuple, Type, Union
+from typing import Any, Callable, Dict, IO, List, Optional, Tuple, Type, Union
 
 from unionml_amd import type_guards
 from unionml_amd._logging import logger
