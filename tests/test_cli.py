@@ -262,3 +262,14 @@ def test_schedule_lifecycle_via_cli(sched_cli_app):
         name = "retrain" if "training" in cmd else "batch_pred"
         r = runner.invoke(app, [cmd, "sched_cli_app:model", name])
         assert r.exit_code == 0, r.output
+
+
+def test_run_scheduler_cli_bounded(sched_cli_app):
+    r = runner.invoke(app, ["deploy", "sched_cli_app:model", "--allow-uncommitted"])
+    assert r.exit_code == 0, r.output
+    # bounded loop: primes next-fire times and exits cleanly without firing
+    r = runner.invoke(
+        app, ["run-scheduler", "sched_cli_app:model", "-n", "2", "--poll-s", "0.01"]
+    )
+    assert r.exit_code == 0, r.output
+    assert "scheduler running" in r.output
