@@ -40,8 +40,9 @@ def parse_args():
     p.add_argument(
         "--graph-steps",
         type=int,
-        default=16,
-        help="optimizer steps captured per hipGraph (amortizes replay launch)",
+        default=64,
+        help="optimizer steps captured per hipGraph (amortizes replay launch; "
+        "64 = one graph per full minibatch cycle, measured +2%% over 16)",
     )
     p.add_argument(
         "--engine",
