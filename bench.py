@@ -7,7 +7,7 @@ synthetic digits-shaped data (64 features, 10 classes, random-init
 weights — no network for datasets) with the CDNA4 fused hot path:
 one optimizer step = the fully-fused fwd/bwd/reduce/Adam MFMA kernel
 (or, under DP, fused fwd/bwd kernel + RCCL gradient all-reduce + fused
-Adam), with 16 optimizer steps captured per hipGraph so replay launch
+Adam), with 64 optimizer steps captured per hipGraph so replay launch
 overhead is amortized (--graph-steps).
 
 Launch (driver contract):
