@@ -15,7 +15,7 @@ Launch (driver contract):
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
       --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
 
-Weak scaling: per-GPU batch is fixed (default 512); reported value is
+Weak scaling: per-GPU batch is fixed (default 2048); reported value is
 the WHOLE-JOB samples/sec aggregated over all ranks.
 """
 
