@@ -35,6 +35,7 @@ def parse_args():
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--bucket-mb", type=float, default=16.0, help="gradient bucket size; ResNet-18 grads are ~47MB fp32, so 16MB gives ~3 buckets overlapping backward")
     p.add_argument("--minibatches", type=int, default=4)
+    p.add_argument("--no-graph", action="store_true")
     return p.parse_args()
 
 
@@ -74,7 +75,12 @@ def main():
         if world > 1
         else None
     )
-    opt = torch.optim.Adam(net.parameters(), lr=args.lr, foreach=True)
+    want_graphs = use_gpu and not args.no_graph
+    # capturable must be set BEFORE the first opt.step so Adam keeps its
+    # step counters on-device (required for hipGraph capture)
+    opt = torch.optim.Adam(
+        net.parameters(), lr=args.lr, foreach=True, capturable=want_graphs
+    )
 
     B, M = args.batch, args.minibatches
     gen = torch.Generator(device="cpu").manual_seed(100 + rank)
@@ -111,11 +117,37 @@ def main():
     for k in range(args.warmup):
         loss = step(k)
     barrier_sync()
+
+    # capture the whole train step (fwd+bwd+reduce+Adam) per minibatch
+    # into hipGraphs — a ResNet-18 step is a few hundred kernel launches
+    engine = "eager"
+    graphs = None
+    if want_graphs:
+        try:
+            graphs = []
+            for mb in range(M):
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    step(mb)
+                graphs.append(g)
+            engine = "hipgraph"
+        except RuntimeError as exc:
+            print(f"[bench_resnet] graph capture unavailable ({exc}); eager",
+                  file=sys.stderr)
+            graphs = None
+    barrier_sync()
+
     t0 = time.perf_counter()
     for k in range(args.steps):
-        loss = step(k)
+        if graphs is not None:
+            graphs[k % M].replay()
+        else:
+            loss = step(k)
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    if graphs is not None:
+        loss = step(args.steps % M)  # fresh eager step for a readable loss
+        barrier_sync()
 
     if dist is not None:
         t = torch.tensor([elapsed], device=device if use_gpu else "cpu")
@@ -148,6 +180,7 @@ def main():
                         "parallelism": f"dp{max(world, 1)}",
                         "image_size": args.image_size,
                         "bucket_mb": args.bucket_mb,
+                        "engine": engine,
                         "final_loss": final_loss,
                     },
                 }
