@@ -35,7 +35,12 @@ def parse_args():
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--bucket-mb", type=float, default=16.0, help="gradient bucket size; ResNet-18 grads are ~47MB fp32, so 16MB gives ~3 buckets overlapping backward")
     p.add_argument("--minibatches", type=int, default=4)
-    p.add_argument("--no-graph", action="store_true")
+    p.add_argument(
+        "--graph",
+        action="store_true",
+        help="EXPERIMENTAL: capture the whole train step into hipGraphs "
+        "(full-autograd capture segfaults on this stack — default off)",
+    )
     return p.parse_args()
 
 
@@ -75,7 +80,7 @@ def main():
         if world > 1
         else None
     )
-    want_graphs = use_gpu and not args.no_graph
+    want_graphs = use_gpu and args.graph
     # capturable must be set BEFORE the first opt.step so Adam keeps its
     # step counters on-device (required for hipGraph capture)
     opt = torch.optim.Adam(
