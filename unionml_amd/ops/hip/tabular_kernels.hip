@@ -764,15 +764,37 @@ adam_step_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
     corr2 = fast_rcp(1.f - __powf(beta2, (float)t));
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < NPARAM; i += 256) {
-    const float g = grads[i];
-    const float mi = beta1 * m[i] + (1.f - beta1) * g;
-    const float vi = beta2 * v[i] + (1.f - beta2) * g * g;
-    m[i] = mi;
-    v[i] = vi;
-    const float p = master[i] - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
-    master[i] = p;
-    bfmirror[i] = f2bf(p);
+  // batch 4 strided elements' loads per iteration so the g/m/v/master
+  // reads overlap instead of serializing one HBM round trip per element
+  // (the serial version measured 6.7 µs for 2.6k params — pure latency)
+  for (int base = (int)threadIdx.x; base < NPARAM; base += 256 * 4) {
+    float g[4], mi_[4], vi_[4], p_[4];
+    int idx[4], n = 0;
+    #pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int i = base + u * 256;
+      if (i < NPARAM) {
+        idx[n] = i;
+        g[n] = grads[i];
+        mi_[n] = m[i];
+        vi_[n] = v[i];
+        p_[n] = master[i];
+        ++n;
+      }
+    }
+    #pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      if (u < n) {
+        const float mi = beta1 * mi_[u] + (1.f - beta1) * g[u];
+        const float vi = beta2 * vi_[u] + (1.f - beta2) * g[u] * g[u];
+        const int i = idx[u];
+        m[i] = mi;
+        v[i] = vi;
+        const float p = p_[u] - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
+        master[i] = p;
+        bfmirror[i] = f2bf(p);
+      }
+    }
   }
 }
 
