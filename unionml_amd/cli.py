@@ -39,6 +39,17 @@ app = typer.Typer(
     no_args_is_help=True,
 )
 
+
+@app.callback(invoke_without_command=True)
+def _main_callback(
+    version: bool = typer.Option(False, "--version", help="print version and exit"),
+):
+    if version:
+        import unionml_amd
+
+        typer.echo(f"unionml-amd {unionml_amd.__version__}")
+        raise typer.Exit()
+
 TEMPLATES_DIR = Path(__file__).parent / "templates"
 
 
