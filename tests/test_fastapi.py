@@ -355,3 +355,27 @@ def test_reload_without_source_400():
     os.environ.pop("UNIONML_MODEL_PATH", None)
     with TestClient(app) as client:
         assert client.post("/reload").status_code == 400
+
+
+def test_predict_malformed_inputs_never_crash(trained_app):
+    """Error paths: malformed bodies must map to 4xx/5xx JSON errors,
+    never an unhandled exception."""
+    _, app = trained_app
+    with TestClient(app) as client:
+        cases = [
+            ("not json at all", None),                      # invalid JSON
+            (None, [1, 2, 3]),                              # non-object body
+            (None, {"features": "not-a-record-list"}),      # wrong features type
+            (None, {"features": [{"wrong": 1.0}]}),         # missing columns
+            (None, {"inputs": {"nonexistent_kwarg": 1}}),   # bad reader kwarg
+            (None, {"features": []}),                       # empty list
+        ]
+        for raw, body in cases:
+            if raw is not None:
+                resp = client.post(
+                    "/predict", content=raw, headers={"Content-Type": "application/json"}
+                )
+            else:
+                resp = client.post("/predict", json=body)
+            assert resp.status_code in (400, 422, 500), (raw, body, resp.status_code)
+            assert resp.json().get("detail"), (raw, body)
