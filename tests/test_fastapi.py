@@ -379,3 +379,18 @@ def test_predict_malformed_inputs_never_crash(trained_app):
                 resp = client.post("/predict", json=body)
             assert resp.status_code in (400, 422, 500), (raw, body, resp.status_code)
             assert resp.json().get("detail"), (raw, body)
+
+
+def test_batcher_route_propagates_errors_cleanly():
+    """A bad request through the batcher must 500 with detail and leave
+    the batcher serving subsequent good requests."""
+    model = build_sklearn_app()
+    model.train()
+    app = FastAPI()
+    model.serve(app, batch=True)
+    good = [{"x1": 0.5, "x2": 0.1, "x3": 0.9}]
+    with TestClient(app) as client:
+        r = client.post("/predict", json={"features": [{"bogus": 1}]})
+        assert r.status_code == 500 and r.json().get("detail")
+        r = client.post("/predict", json={"features": good})
+        assert r.status_code == 200, r.text
