@@ -188,3 +188,29 @@ def test_dataset_from_task():
     (Xtr, ytr), (Xte, yte) = data["train"], data["test"]
     assert len(Xtr) + len(Xte) == 20
     assert list(Xtr.columns) == ["a"]
+
+
+def test_from_sqlalchemy_task_sqlite_fallback(tmp_path):
+    """from_sqlalchemy_task works with or without sqlalchemy installed
+    (falls back to sqlite3 for sqlite:/// URIs — reference capability:
+    dataset.py:458-470)."""
+    import sqlite3
+
+    import pandas as pd
+
+    from unionml_amd.dataset import Dataset
+
+    db = tmp_path / "t.db"
+    with sqlite3.connect(db) as conn:
+        conn.execute("CREATE TABLE pts (a REAL, y INTEGER)")
+        conn.executemany(
+            "INSERT INTO pts VALUES (?, ?)", [(i * 0.1, i % 2) for i in range(30)]
+        )
+    ds = Dataset.from_sqlalchemy_task(
+        "sql_ds", f"sqlite:///{db}", "SELECT * FROM pts WHERE a < {amax}",
+        targets=["y"], test_size=0.2, random_state=0,
+    )
+    raw = ds.dataset_task()(amax=2.0)
+    assert isinstance(raw, pd.DataFrame) and len(raw) == 20
+    data = ds.get_data(raw)
+    assert len(data["train"][0]) + len(data["test"][0]) == 20
