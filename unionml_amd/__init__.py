@@ -9,9 +9,10 @@ across the 8 GPUs of a node with gradient all-reduce on RCCL over xGMI.
 """
 
 from unionml_amd.dataset import Dataset
+from unionml_amd.defaults import Resources
 from unionml_amd.model import Model, ModelArtifact, BaseHyperparameters
 from unionml_amd.schedule import Schedule
 
-__all__ = ["Dataset", "Model", "ModelArtifact", "BaseHyperparameters", "Schedule"]
+__all__ = ["Dataset", "Model", "ModelArtifact", "BaseHyperparameters", "Schedule", "Resources"]
 
 __version__ = "0.1.0"
