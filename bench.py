@@ -147,7 +147,7 @@ def main():
                     Xbf[off : off + B], y[off : off + B], clf.W1bf, clf.W2bf,
                     clf.master, clf.bfmirror, clf.m, clf.v, clf.t_dev,
                     clf.slabs, clf.counter, loss_out, invBtot,
-                    args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                    args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=clf.wimg,
                 )
                 assert ok
 
@@ -163,13 +163,14 @@ def main():
                     clf.master, clf.bfmirror, clf.m, clf.v, clf.t_dev,
                     clf.slabs, clf.counter, loss_out, invBtot,
                     args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
-                    grads_out=clf.grads,
+                    grads_out=clf.grads, wimg=clf.wimg,
                 )
                 assert ok
                 if dist is not None:
                     dist.all_reduce(clf.grads)
                 ext.adam_step(clf.master, clf.bfmirror, clf.grads, clf.m, clf.v,
-                              clf.t_dev, args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS)
+                              clf.t_dev, args.lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                              wimg=clf.wimg)
 
     else:
 

@@ -295,3 +295,40 @@ def test_step_fused_reduce_only_matches_mlp_step(ext, dev):
     err = (a.grads[: ref.NPARAM + 1] - b.grads[: ref.NPARAM + 1]).abs().max().item()
     scale = a.grads.abs().max().item()
     assert err < max(1e-5 * scale, 1e-6), (err, scale)
+
+
+def test_wimg_path_matches_plain(ext, dev):
+    """The packed weight-image path (wimg=) must train identically to
+    the transposed-gather path, across fused and stepwise flows."""
+    from unionml_amd.ops import reference as ref
+    from unionml_amd.ops.tabular import ADAM_BETA1, ADAM_BETA2, ADAM_EPS, TabularMLP
+
+    torch.manual_seed(41)
+    B, n_steps = 512, 9
+    Xbf = (torch.randn(B, 64) * 1.2 + 0.1).bfloat16().to(dev)
+    y = torch.randint(0, 10, (B,), dtype=torch.int32, device=dev)
+
+    def train(use_wimg):
+        c = TabularMLP(device=dev, seed=12)
+        c._ensure_slabs((B + 127) // 128)
+        loss_out = c.grads[ref.NPARAM : ref.NPARAM + 1]
+        for _ in range(n_steps):
+            ok = ext.mlp_step_fused(
+                Xbf, y, c.W1bf, c.W2bf, c.master, c.bfmirror, c.m, c.v,
+                c.t_dev, c.slabs, c.counter, loss_out, 1.0 / B,
+                1e-3, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                wimg=c.wimg if use_wimg else None,
+            )
+            assert ok
+        torch.cuda.synchronize()
+        return c
+
+    a, b = train(False), train(True)
+    err = (a.master - b.master).abs().max().item()
+    assert err < 1e-5, f"wimg path diverged: {err}"
+    # wimg stays in sync with bfmirror after kernel updates
+    rebuilt = TabularMLP(device=dev, seed=12)
+    rebuilt.master.copy_(b.master)
+    rebuilt.bfmirror.copy_(b.master.bfloat16())
+    rebuilt._build_wimg()
+    assert torch.equal(rebuilt.wimg, b.wimg), "wimg out of sync with master"

@@ -261,6 +261,38 @@ __device__ __forceinline__ Lds carve(char* smem) {
   return L;
 }
 
+// packed weight-image staging: a persistent global buffer holding the
+// THREE LDS images (W1T/W2s/W2T incl. their K-pads) back to back in
+// exactly the LDS layout, so the training prologue is one straight
+// vectorized copy instead of a per-element transposed gather. Writers:
+// the Adam phases (in-kernel and adam_step) and the host on
+// init/load_state_dict.
+#define WIMG_W1T 0
+#define WIMG_W2S (HID * XS)
+#define WIMG_W2T (HID * XS + HID * WS)
+#define WIMG_N   (HID * XS + HID * WS + CPAD * WS)
+
+__device__ __forceinline__ void load_weight_images_packed(
+    const Lds& L, const u16* __restrict__ wimg) {
+  bf16x8* dst = (bf16x8*)&L.W1T[0][0];   // W1T..W2T are contiguous in LDS
+  const bf16x8* src = (const bf16x8*)wimg;
+  for (int i = threadIdx.x; i < WIMG_N / 8; i += BLOCK) dst[i] = src[i];
+}
+
+__device__ __forceinline__ void wimg_write(u16* __restrict__ wimg, int i, u16 wb) {
+  if (i < OFF_B1) {                       // W1: i = in*HID + h
+    const int in = i >> 5;                // /HID (32)
+    const int h = i & 31;
+    wimg[WIMG_W1T + h * XS + in] = wb;
+  } else if (i >= OFF_W2 && i < OFF_B2) { // W2: j = h*CPAD + c
+    const int j = i - OFF_W2;
+    const int h = j >> 4;                 // /CPAD (16)
+    const int c = j & 15;
+    wimg[WIMG_W2S + h * WS + c] = wb;
+    wimg[WIMG_W2T + c * WS + h] = wb;
+  }
+}
+
 // prefetch biases into LDS alongside the other prologue loads — the
 // previous launch's agent-scope acquire left L2 cold, so a lazy mid-GEMM
 // b1/b2 read would serialize a full HBM round trip into the fwd chain
@@ -558,10 +590,11 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
                       float* __restrict__ loss_out,
                       float invBtot, float lr, float beta1, float beta2,
                       float eps,
-                      float* __restrict__ grads_out) { // non-null: write summed
+                      float* __restrict__ grads_out, // non-null: write summed
                                                        // grads (+loss) and skip
                                                        // Adam — the DP path's
                                                        // pre-collective kernel
+                      u16* __restrict__ wimg) {        // optional packed images
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const Lds L = carve(smem);
   // L.loss slots: [0] loss accum, [1] (bits) poll base, [2] t_pre, [3] unused
@@ -587,7 +620,8 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
   }
   zero_dl_pad(L);
 #ifndef PROBE_SKIP_LOADS
-  load_weight_images(L, W1bf, W2bf);
+  if (wimg) load_weight_images_packed(L, wimg);
+  else load_weight_images(L, W1bf, W2bf);
   load_x_chunk(Xbf, L, row0, B);
 #endif
   load_biases(L, master + OFF_B1, master + OFF_B2);
@@ -740,7 +774,9 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
     v[i] = vi;
     const float p = p_old - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
     master[i] = p;
-    bfmirror[i] = f2bf(p);
+    const u16 wb = f2bf(p);
+    bfmirror[i] = wb;
+    if (wimg) wimg_write(wimg, i, wb);
   }
   if (tid == 0 && blockIdx.x == 0) {
     if (grads_out) *counter = epoch;
@@ -756,7 +792,8 @@ extern "C" __global__ void __launch_bounds__(256)
 adam_step_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
                  const float* __restrict__ grads, float* __restrict__ m,
                  float* __restrict__ v, int* __restrict__ t_dev,
-                 float lr, float beta1, float beta2, float eps) {
+                 float lr, float beta1, float beta2, float eps,
+                 u16* __restrict__ wimg) {
   __shared__ float corr1, corr2;
   if (threadIdx.x == 0) {
     const int t = ++(*t_dev);
@@ -792,7 +829,9 @@ adam_step_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
         v[i] = vi;
         const float p = p_[u] - lr * (mi * corr1) * fast_rcp(sqrtf(vi * corr2) + eps);
         master[i] = p;
-        bfmirror[i] = f2bf(p);
+        const u16 wb = f2bf(p);
+        bfmirror[i] = wb;
+        if (wimg) wimg_write(wimg, i, wb);
       }
     }
   }
@@ -1100,7 +1139,7 @@ int launch_mlp_step_fused(const unsigned short* Xbf, const int* y, int B,
                           float* v, int* t_dev, float* slabs, unsigned* counter,
                           float* loss_out, float invBtot, float lr, float beta1,
                           float beta2, float eps, int max_slabs, float* grads_out,
-                          hipStream_t stream) {
+                          unsigned short* wimg, hipStream_t stream) {
   const int blocks = (B + ROWS - 1) / ROWS;
   if (blocks > max_slabs) return -1;
   static int done = 0;
@@ -1111,7 +1150,7 @@ int launch_mlp_step_fused(const unsigned short* Xbf, const int* y, int B,
   hipLaunchKernelGGL(mlp_step_fused_kernel, dim3(blocks), dim3(BLOCK), C_IMG_TOTAL,
                      stream, Xbf, y, B, W1bf, W2bf, master, bfmirror, m, v, t_dev,
                      slabs, counter, loss_out, invBtot, lr, beta1, beta2, eps,
-                     grads_out);
+                     grads_out, wimg);
   return 0;
 }
 
@@ -1145,9 +1184,11 @@ void launch_mlp_predict(const float* X, int B, const float* mean,
 
 void launch_adam_step(float* master, unsigned short* bfmirror, const float* grads,
                       float* m, float* v, int* t_dev, float lr, float beta1,
-                      float beta2, float eps, hipStream_t stream) {
+                      float beta2, float eps, unsigned short* wimg,
+                      hipStream_t stream) {
   hipLaunchKernelGGL(adam_step_kernel, dim3(1), dim3(256), 0, stream,
-                     master, bfmirror, grads, m, v, t_dev, lr, beta1, beta2, eps);
+                     master, bfmirror, grads, m, v, t_dev, lr, beta1, beta2, eps,
+                     wimg);
 }
 
 }  // extern "C"

@@ -20,12 +20,12 @@ int launch_mlp_step_fused(const unsigned short*, const int*, int,
                           const unsigned short*, const unsigned short*, float*,
                           unsigned short*, float*, float*, int*, float*, unsigned*,
                           float*, float, float, float, float, float, int, float*,
-                          hipStream_t);
+                          unsigned short*, hipStream_t);
 void launch_mlp_predict(const float*, int, const float*, const float*,
                         const unsigned short*, const unsigned short*, const float*,
                         int*, float*, hipStream_t);
 void launch_adam_step(float*, unsigned short*, const float*, float*, float*, int*,
-                      float, float, float, float, hipStream_t);
+                      float, float, float, float, unsigned short*, hipStream_t);
 }
 
 namespace {
@@ -101,7 +101,8 @@ bool mlp_step_fused(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
                     torch::Tensor slabs, torch::Tensor counter,
                     torch::Tensor loss_out, double invBtot, double lr,
                     double beta1, double beta2, double eps,
-                    c10::optional<torch::Tensor> grads_out = c10::nullopt) {
+                    c10::optional<torch::Tensor> grads_out = c10::nullopt,
+                    c10::optional<torch::Tensor> wimg = c10::nullopt) {
   check(Xbf, torch::kBFloat16, "Xbf");
   check(y, torch::kInt32, "y");
   check(master, torch::kFloat32, "master");
@@ -117,13 +118,20 @@ bool mlp_step_fused(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
     TORCH_CHECK(grads_out->numel() >= 2609, "grads_out must hold params + loss");
     grads_ptr = grads_out->data_ptr<float>();
   }
+  unsigned short* wimg_ptr = nullptr;
+  if (wimg.has_value()) {
+    check(*wimg, torch::kBFloat16, "wimg");
+    TORCH_CHECK(wimg->numel() == 4224, "wimg must be the packed 4224-elem image buffer");
+    wimg_ptr = bf16_mut_ptr(*wimg);
+  }
   const int rc = launch_mlp_step_fused(
       bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0), bf16_ptr(W1bf),
       bf16_ptr(W2bf), master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
       m.data_ptr<float>(), v.data_ptr<float>(), t_dev.data_ptr<int>(),
       slabs.data_ptr<float>(), (unsigned*)counter.data_ptr(),
       loss_out.data_ptr<float>(), (float)invBtot, (float)lr, (float)beta1,
-      (float)beta2, (float)eps, (int)slabs.size(0), grads_ptr, current_stream());
+      (float)beta2, (float)eps, (int)slabs.size(0), grads_ptr, wimg_ptr,
+      current_stream());
   TORCH_CHECK(rc != -2, "mlp_step_fused: hipFuncSetAttribute(LDS) failed");
   return rc == 0;
 }
@@ -170,17 +178,24 @@ void mlp_predict(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd,
 
 void adam_step(torch::Tensor master, torch::Tensor bfmirror, torch::Tensor grads,
                torch::Tensor m, torch::Tensor v, torch::Tensor t_dev, double lr,
-               double beta1, double beta2, double eps) {
+               double beta1, double beta2, double eps,
+               c10::optional<torch::Tensor> wimg = c10::nullopt) {
   check(master, torch::kFloat32, "master");
   check(bfmirror, torch::kBFloat16, "bfmirror");
   check(grads, torch::kFloat32, "grads");
   check(m, torch::kFloat32, "m");
   check(v, torch::kFloat32, "v");
   check(t_dev, torch::kInt32, "t_dev");
+  unsigned short* wimg_ptr = nullptr;
+  if (wimg.has_value()) {
+    check(*wimg, torch::kBFloat16, "wimg");
+    TORCH_CHECK(wimg->numel() == 4224, "wimg must be the packed 4224-elem image buffer");
+    wimg_ptr = bf16_mut_ptr(*wimg);
+  }
   launch_adam_step(master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
                    grads.data_ptr<float>(), m.data_ptr<float>(), v.data_ptr<float>(),
                    t_dev.data_ptr<int>(), (float)lr, (float)beta1, (float)beta2,
-                   (float)eps, current_stream());
+                   (float)eps, wimg_ptr, current_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -194,9 +209,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("master"), py::arg("bfmirror"), py::arg("m"), py::arg("v"),
         py::arg("t_dev"), py::arg("slabs"), py::arg("counter"),
         py::arg("loss_out"), py::arg("invBtot"), py::arg("lr"), py::arg("beta1"),
-        py::arg("beta2"), py::arg("eps"), py::arg("grads_out") = c10::nullopt);
+        py::arg("beta2"), py::arg("eps"), py::arg("grads_out") = c10::nullopt,
+        py::arg("wimg") = c10::nullopt);
   m.def("mlp_train_steps", &mlp_train_steps,
         "persistent multi-step training kernel (weights+Adam resident in LDS)");
   m.def("mlp_predict", &mlp_predict, "fused standardize+fwd+argmax (CDNA4 MFMA)");
-  m.def("adam_step", &adam_step, "fused Adam on flat master params (CDNA4)");
+  m.def("adam_step", &adam_step, "fused Adam on flat master params (CDNA4)",
+        py::arg("master"), py::arg("bfmirror"), py::arg("grads"), py::arg("m"),
+        py::arg("v"), py::arg("t_dev"), py::arg("lr"), py::arg("beta1"),
+        py::arg("beta2"), py::arg("eps"), py::arg("wimg") = c10::nullopt);
 }

@@ -56,6 +56,29 @@ class TabularMLP:
         self.counter = None     # G16 ticket counter (self-resetting)
         self._graph = None
         self._graph_key = None
+        # packed weight images (the three LDS layouts back to back) so
+        # the training prologue is a straight vectorized copy; kept in
+        # sync by the Adam kernels (wimg=) and rebuilt on any host-side
+        # weight mutation
+        self.wimg = (
+            torch.zeros(4224, dtype=torch.bfloat16, device=self.device)
+            if self.use_hip
+            else None
+        )
+        self._build_wimg()
+
+    _XS, _WS = 72, 40  # LDS row strides (tabular_kernels.hip)
+
+    def _build_wimg(self):
+        if self.wimg is None:
+            return
+        XS, WS = self._XS, self._WS
+        W1, _, W2, _ = ref.unpack_master(self.master.detach().cpu())
+        img = torch.zeros(4224, dtype=torch.bfloat16)
+        img[: HID * XS].view(HID, XS)[:, :IN] = W1.t().bfloat16()      # W1T[h][k]
+        img[HID * XS : HID * XS + HID * WS].view(HID, WS)[:, :CPAD] = W2.bfloat16()  # W2s[h][c]
+        img[HID * XS + HID * WS :].view(CPAD, WS)[:, :HID] = W2.t().bfloat16()       # W2T[c][h]
+        self.wimg.copy_(img.to(self.device))
 
     def _ensure_slabs(self, n_wg: int):
         if self.slabs is None or self.slabs.shape[0] < n_wg:
@@ -105,7 +128,7 @@ class TabularMLP:
                 Xbf, y, self.W1bf, self.W2bf, self.master, self.bfmirror,
                 self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
                 invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
-                grads_out=self.grads,
+                grads_out=self.grads, wimg=self.wimg,
             )
             assert ok, "mlp_step_fused slab capacity exceeded"
         else:
@@ -118,7 +141,7 @@ class TabularMLP:
         if self.use_hip:
             hip_ext().adam_step(
                 self.master, self.bfmirror, self.grads, self.m, self.v, self.t_dev,
-                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
             )
         else:
             self.t_dev += 1
@@ -139,7 +162,7 @@ class TabularMLP:
                     Xbf[off : off + bs], y[off : off + bs], self.W1bf, self.W2bf,
                     self.master, self.bfmirror, self.m, self.v, self.t_dev,
                     self.slabs, self.counter, loss_out, 1.0 / bs,
-                    lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                    lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
                 )
                 assert ok, "mlp_step_fused slab capacity exceeded"
 
@@ -279,6 +302,7 @@ class TabularMLP:
         d = dict(self.__dict__)
         d["slabs"] = d["counter"] = None
         d["_graph"] = d["_graph_key"] = None
+        d["wimg"] = None
         d["device"] = str(self.device)
         for k, v in list(d.items()):
             if torch.is_tensor(v):
@@ -297,6 +321,9 @@ class TabularMLP:
         for k, v in list(self.__dict__.items()):
             if torch.is_tensor(v):
                 setattr(self, k, v.to(device))
+        if self.use_hip and self.wimg is None:
+            self.wimg = torch.zeros(4224, dtype=torch.bfloat16, device=device)
+        self._build_wimg()
 
     def state_dict(self) -> Dict[str, torch.Tensor]:
         W1, b1, W2, b2 = ref.unpack_master(self.master.cpu())
@@ -321,3 +348,4 @@ class TabularMLP:
         self.mean.copy_(state["mean"].to(self.device))
         self.invstd.copy_(state["invstd"].to(self.device))
         self._graph = self._graph_key = None
+        self._build_wimg()
