@@ -23,6 +23,7 @@ int main(int argc, char** argv) {
   int *y, *t_dev;
   unsigned* counter;
   float *master, *m, *v, *slabs, *loss;
+  unsigned short* wimg;
   CHECK(hipMalloc(&Xbf, (size_t)B * 64 * 2));
   CHECK(hipMalloc(&W1bf, 64 * 32 * 2));
   CHECK(hipMalloc(&W2bf, 32 * 16 * 2));
@@ -35,6 +36,8 @@ int main(int argc, char** argv) {
   CHECK(hipMalloc(&v, NPARAM * 4));
   CHECK(hipMalloc(&slabs, (size_t)n_wg * SLAB * 4));
   CHECK(hipMalloc(&loss, 4));
+  CHECK(hipMalloc(&wimg, 4224 * 2));
+  CHECK(hipMemset(wimg, 0, 4224 * 2));
   CHECK(hipMemset(Xbf, 0, (size_t)B * 64 * 2));
   CHECK(hipMemset(W1bf, 0, 64 * 32 * 2));
   CHECK(hipMemset(W2bf, 0, 32 * 16 * 2));
@@ -49,7 +52,7 @@ int main(int argc, char** argv) {
   for (int i = 0; i < 100; ++i) {
     if (launch_mlp_step_fused(Xbf, y, B, W1bf, W2bf, master, bfmirror, m, v,
                               t_dev, slabs, counter, loss, 1.0f / B, 1e-3f,
-                              0.9f, 0.999f, 1e-8f, n_wg, nullptr, 0) != 0) {
+                              0.9f, 0.999f, 1e-8f, n_wg, nullptr, wimg, 0) != 0) {
       fprintf(stderr, "launch failed\n");
       return 1;
     }
@@ -63,7 +66,7 @@ int main(int argc, char** argv) {
   for (int i = 0; i < iters; ++i) {
     launch_mlp_step_fused(Xbf, y, B, W1bf, W2bf, master, bfmirror, m, v, t_dev,
                           slabs, counter, loss, 1.0f / B, 1e-3f, 0.9f, 0.999f,
-                          1e-8f, n_wg, nullptr, 0);
+                          1e-8f, n_wg, nullptr, wimg, 0);
   }
   CHECK(hipEventRecord(t1, 0));
   CHECK(hipEventSynchronize(t1));
