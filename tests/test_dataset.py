@@ -214,3 +214,29 @@ def test_from_sqlalchemy_task_sqlite_fallback(tmp_path):
     assert isinstance(raw, pd.DataFrame) and len(raw) == 20
     data = ds.get_data(raw)
     assert len(data["train"][0]) + len(data["test"][0]) == 20
+
+
+def test_default_loader_rejects_loader_kwargs(frame_dataset):
+    """loader_kwargs with the default loader is a GuardError, not a
+    TypeError from the default loader's signature."""
+    from unionml_amd.type_guards import GuardError
+
+    raw = frame_dataset.dataset_task()(n=10)
+    with pytest.raises(GuardError, match="no @dataset.loader"):
+        frame_dataset.get_data(raw, loader_kwargs={"anything": 1})
+
+
+def test_feature_loader_fast_path_preserves_dtypes():
+    """The /predict records fast path must keep per-column dtypes (ints
+    stay ints, bools stay bools) — not coerce everything to float64."""
+    ds = Dataset(name="dtypes", features=["a", "b", "c"], targets=["t"])
+
+    @ds.reader
+    def reader() -> pd.DataFrame:
+        return pd.DataFrame({"a": [1], "b": [0.5], "c": [True], "t": [0]})
+
+    feats = ds.get_features([{"a": 1, "b": 0.25, "c": True}, {"a": 2, "b": 0.75, "c": False}])
+    assert feats["a"].dtype.kind == "i"
+    assert feats["b"].dtype.kind == "f"
+    assert feats["c"].dtype.kind == "b"
+    assert feats["a"].tolist() == [1, 2]

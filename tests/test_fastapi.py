@@ -394,3 +394,22 @@ def test_batcher_route_propagates_errors_cleanly():
         assert r.status_code == 500 and r.json().get("detail")
         r = client.post("/predict", json={"features": good})
         assert r.status_code == 200, r.text
+
+
+def test_reload_token_gate(tmp_path, monkeypatch):
+    """With $UNIONML_RELOAD_TOKEN set, /reload requires the matching
+    X-Reload-Token header (advisor finding: unauthenticated reloads can
+    force artifact swaps + hipGraph re-capture churn)."""
+    model = build_sklearn_app()
+    model.train()
+    path = tmp_path / "m.joblib"
+    model.save(path)
+
+    app = FastAPI()
+    model.serve(app)
+    monkeypatch.setenv("UNIONML_MODEL_PATH", str(path))
+    monkeypatch.setenv("UNIONML_RELOAD_TOKEN", "s3cret")
+    with TestClient(app) as client:
+        assert client.post("/reload").status_code == 403
+        assert client.post("/reload", headers={"X-Reload-Token": "wrong"}).status_code == 403
+        assert client.post("/reload", headers={"X-Reload-Token": "s3cret"}).status_code == 200

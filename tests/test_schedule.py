@@ -79,3 +79,44 @@ def test_next_fire_time_fixed_rate():
     s = Schedule(type="trainer", name="s", fixed_rate=datetime.timedelta(hours=2))
     after = datetime.datetime(2026, 9, 13, 2, 0)
     assert next_fire_time(s, after) == datetime.datetime(2026, 9, 13, 4, 0)
+
+
+def test_parse_iso_duration():
+    from unionml_amd.schedule import parse_iso_duration
+
+    assert parse_iso_duration("PT10M") == datetime.timedelta(minutes=10)
+    assert parse_iso_duration("P2DT3H30M15S") == datetime.timedelta(
+        days=2, hours=3, minutes=30, seconds=15
+    )
+    assert parse_iso_duration("P1W") == datetime.timedelta(weeks=1)
+    assert parse_iso_duration("-PT5M") == -datetime.timedelta(minutes=5)
+    for bad in ("", "P", "PT", "10M", "P1M2W", "nonsense"):
+        with pytest.raises(ValueError):
+            parse_iso_duration(bad)
+
+
+def test_next_fire_time_cron_offset():
+    """A cron offset shifts every kickoff by the duration (reference
+    schedule.py:99-103 passes offset into the platform CronSchedule)."""
+    s = Schedule(type="trainer", name="s", expression="30 4 * * *", offset="PT15M")
+    after = datetime.datetime(2026, 9, 13, 2, 0)
+    assert next_fire_time(s, after) == datetime.datetime(2026, 9, 13, 4, 45)
+    # strictly-after holds across the offset boundary: at 04:40 the
+    # 04:30 match (firing 04:45) is still pending
+    after2 = datetime.datetime(2026, 9, 13, 4, 40)
+    assert next_fire_time(s, after2) == datetime.datetime(2026, 9, 13, 4, 45)
+    # past 04:45 we roll to the next day
+    after3 = datetime.datetime(2026, 9, 13, 4, 50)
+    assert next_fire_time(s, after3) == datetime.datetime(2026, 9, 14, 4, 45)
+
+
+def test_negative_offset():
+    s = Schedule(type="trainer", name="s", expression="0 5 * * *", offset="-PT30M")
+    after = datetime.datetime(2026, 9, 13, 2, 0)
+    assert next_fire_time(s, after) == datetime.datetime(2026, 9, 13, 4, 30)
+
+
+def test_invalid_offset_rejected_at_deploy():
+    s = Schedule(type="trainer", name="s", expression="0 5 * * *", offset="15 minutes")
+    with pytest.raises(ValueError):
+        create_scheduled_launchplan("m.train", "s", s)

@@ -246,6 +246,13 @@ class Dataset(TrackedInstance):
         splitter_kwargs = {**self.splitter_kwargs, **(splitter_kwargs or {})}
         parser_kwargs = {**self.parser_kwargs, **(parser_kwargs or {})}
 
+        if loader_kwargs and self._loader == self._default_loader:
+            from unionml_amd.type_guards import GuardError
+
+            raise GuardError(
+                "loader_kwargs were provided but no @dataset.loader is registered: "
+                "the default loader takes no keyword arguments"
+            )
         data = self._loader(raw_data, **loader_kwargs)
         splits = self._splitter(data, **splitter_kwargs)
 
@@ -399,9 +406,12 @@ class Dataset(TrackedInstance):
         if data_type is pd.DataFrame:
             feature_names = self._features
             # serve-time fast path: list of records with declared feature
-            # columns -> build the ndarray directly in column order (an
-            # order of magnitude cheaper than pandas' per-record
-            # inference; this sits on the /predict hot path)
+            # columns -> build one ndarray per column (an order of
+            # magnitude cheaper than pandas' per-record inference; this
+            # sits on the /predict hot path). Per-column arrays keep each
+            # column's dtype (ints stay ints, bools stay bools) — a
+            # single float64 matrix would silently change dtypes for
+            # dtype-sensitive models.
             if (
                 feature_names
                 and isinstance(features, list)
@@ -409,13 +419,11 @@ class Dataset(TrackedInstance):
                 and isinstance(features[0], dict)
             ):
                 try:
-                    arr = np.asarray(
-                        [[rec[c] for c in feature_names] for rec in features],
-                        dtype=np.float64,
-                    )
-                    return pd.DataFrame(arr, columns=feature_names)
+                    cols = {c: np.asarray([rec[c] for rec in features]) for c in feature_names}
+                    if all(v.dtype != object for v in cols.values()):
+                        return pd.DataFrame(cols, columns=feature_names)
                 except (KeyError, TypeError, ValueError):
-                    pass  # missing keys / non-numeric values: general path
+                    pass  # missing keys / ragged values: general path
             data = pd.DataFrame(features)
             if not feature_names and self._targets:
                 feature_names = [col for col in data.columns if col not in self._targets]

@@ -133,11 +133,20 @@ def serving_app(
         return {"status": "ok"}
 
     @app.post("/reload")
-    async def reload_model():
+    async def reload_model(request: Request):
         """Hot-reload the artifact (after e.g. `unionml-amd fetch-model`)
         without restarting the server: re-read $UNIONML_MODEL_PATH (or
         the backend's latest), swap the artifact, and rebuild the
-        batcher's hipGraphs against the new weights."""
+        batcher's hipGraphs against the new weights.
+
+        Reloads force GPU/memory churn (full hipGraph re-capture), so
+        when $UNIONML_RELOAD_TOKEN is set the request must carry it in
+        the X-Reload-Token header; unauthenticated reloads are only
+        allowed when no token was configured (trusted-network default).
+        """
+        token = os.environ.get("UNIONML_RELOAD_TOKEN")
+        if token and request.headers.get("x-reload-token") != token:
+            raise HTTPException(status_code=403, detail="invalid or missing X-Reload-Token")
         model_path = os.environ.get("UNIONML_MODEL_PATH")
         try:
             if model_path:

@@ -225,6 +225,10 @@ class TabularMLP:
                 )
                 if ok:
                     torch.cuda.synchronize(self.device)
+                    # the persistent kernel updates master/bfmirror but not
+                    # the packed weight images; refresh them so a later
+                    # fused/adam step (wimg=) doesn't load stale weights
+                    self._build_wimg()
                     return float(loss_out.item())
             return self._train_epochs_fused(
                 Xbf, y, batches, epochs=epochs, lr=lr, use_graph=use_graph

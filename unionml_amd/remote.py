@@ -138,12 +138,26 @@ class GpuAllocator:
     def assign(self, n_gpus: int) -> Optional[str]:
         if n_gpus <= 0 or self.n_devices == 0:
             return None
-        try:
-            counter = int(self._counter_file.read_text())
-        except (FileNotFoundError, ValueError):
-            counter = 0
-        devices = [str((counter + i) % self.n_devices) for i in range(min(n_gpus, self.n_devices))]
-        self._counter_file.write_text(str((counter + n_gpus) % self.n_devices))
+        # read-modify-write under an exclusive flock: concurrent execute()
+        # calls (scheduled fan-out across 8 GPUs) must not hand two
+        # workers the same device
+        import fcntl
+
+        self._counter_file.parent.mkdir(parents=True, exist_ok=True)
+        with open(self._counter_file, "a+") as f:
+            fcntl.flock(f, fcntl.LOCK_EX)
+            f.seek(0)
+            try:
+                counter = int(f.read().strip() or 0)
+            except ValueError:
+                counter = 0
+            devices = [
+                str((counter + i) % self.n_devices) for i in range(min(n_gpus, self.n_devices))
+            ]
+            f.seek(0)
+            f.truncate()
+            f.write(str((counter + n_gpus) % self.n_devices))
+            f.flush()
         return ",".join(devices)
 
 

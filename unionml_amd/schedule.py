@@ -90,16 +90,50 @@ def cron_matches(expression: str, when: datetime.datetime) -> bool:
     )
 
 
+def parse_iso_duration(spec: str) -> datetime.timedelta:
+    """Parse an ISO-8601 duration (``P2DT3H30M15S``, ``PT10M``, ``-PT5M``)
+    into a timedelta. Calendar units wider than weeks (months/years) are
+    rejected — they have no fixed length."""
+    import re
+
+    m = re.fullmatch(
+        r"(?P<sign>[-+])?P(?:(?P<weeks>\d+(?:\.\d+)?)W)?(?:(?P<days>\d+(?:\.\d+)?)D)?"
+        r"(?:T(?:(?P<hours>\d+(?:\.\d+)?)H)?(?:(?P<minutes>\d+(?:\.\d+)?)M)?"
+        r"(?:(?P<seconds>\d+(?:\.\d+)?)S)?)?",
+        spec.strip(),
+    )
+    if not m or spec.strip().rstrip("+-") in ("P", "PT", ""):
+        raise ValueError(f"invalid ISO-8601 duration: {spec!r}")
+    parts = {k: float(v) for k, v in m.groupdict().items() if v not in (None, "-", "+")}
+    sign = -1 if m.group("sign") == "-" else 1
+    if not any(k in parts for k in ("weeks", "days", "hours", "minutes", "seconds")):
+        raise ValueError(f"invalid ISO-8601 duration: {spec!r}")
+    return sign * datetime.timedelta(
+        weeks=parts.get("weeks", 0.0),
+        days=parts.get("days", 0.0),
+        hours=parts.get("hours", 0.0),
+        minutes=parts.get("minutes", 0.0),
+        seconds=parts.get("seconds", 0.0),
+    )
+
+
 def next_fire_time(schedule: Schedule, after: datetime.datetime) -> datetime.datetime:
-    """Next time this schedule should fire strictly after ``after``."""
+    """Next time this schedule should fire strictly after ``after``.
+
+    A cron ``offset`` (ISO-8601 duration, reference schedule.py:99-103)
+    shifts each kickoff relative to its cron match: fire times are
+    ``cron_match + offset``.
+    """
     if schedule.fixed_rate is not None:
         return after + schedule.fixed_rate
     if schedule.expression is None:
         raise ValueError(f"schedule '{schedule.name}' has neither expression nor fixed_rate")
-    t = after.replace(second=0, microsecond=0) + datetime.timedelta(minutes=1)
+    offset = parse_iso_duration(schedule.offset) if schedule.offset else datetime.timedelta()
+    base = after - offset
+    t = base.replace(second=0, microsecond=0) + datetime.timedelta(minutes=1)
     for _ in range(60 * 24 * 366):  # search up to ~a year of minutes
         if cron_matches(schedule.expression, t):
-            return t
+            return t + offset
         t += datetime.timedelta(minutes=1)
     raise ValueError(f"cron expression {schedule.expression!r} never fires")
 
@@ -146,6 +180,8 @@ def create_scheduled_launchplan(
         cron_matches(schedule.expression, datetime.datetime.now())
     if schedule.offset is not None and schedule.fixed_rate is not None:
         raise ValueError(f"schedule '{name}': 'offset' only applies to cron expressions")
+    if schedule.offset is not None:
+        parse_iso_duration(schedule.offset)  # fail at deploy time, not fire time
 
     fixed_inputs = dict(schedule.inputs or {})
     fixed_inputs.update(launchplan_kwargs.pop("fixed_inputs", {}) or {})

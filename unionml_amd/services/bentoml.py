@@ -134,8 +134,6 @@ class BentoMLService:
             model_object = self.model.artifact.model_object
         framework = framework or infer_framework(model_object)
         saver = getattr(bentoml, framework)
-        if framework == "pytorch":
-            return saver.save_model(self.name, model_object)
         return saver.save_model(self.name, model_object)
 
     def load_model(self, tag: Optional[str] = None):

@@ -17,8 +17,9 @@ predictor — one code path, two execution substrates.
 import asyncio
 import threading
 import time
+from collections import deque
 from dataclasses import dataclass
-from typing import Any, List, Optional
+from typing import Any, Deque, List, Optional
 
 from unionml_amd._logging import logger
 
@@ -47,14 +48,18 @@ class DynamicBatcher:
         self.model = model
         self.max_batch_size = max_batch_size
         self.max_delay_s = max_delay_ms / 1000.0
-        self._queue: List[_Request] = []
+        self._queue: Deque[_Request] = deque()
         self._cv = threading.Condition()
         self._stop = False
         self._thread: Optional[threading.Thread] = None
         self._graphed = None  # set lazily on first GPU batch
         self._n_batches = 0
-        self._batch_rows: List[int] = []
-        self._done_latency_ms: List[float] = []
+        self._n_requests = 0
+        # bounded reservoirs: a long-running server must not leak one
+        # entry per request (stats() quantiles come from the most recent
+        # window, which is also the operationally interesting one)
+        self._batch_rows: Deque[int] = deque(maxlen=4096)
+        self._done_latency_ms: Deque[float] = deque(maxlen=16384)
 
     def start(self):
         # build the hipGraph runner (and capture all buckets) up front so
@@ -98,7 +103,7 @@ class DynamicBatcher:
         pick = lambda q: lat[min(len(lat) - 1, int(q * len(lat)))] if lat else None  # noqa: E731
         return {
             "batches": self._n_batches,
-            "requests": len(lat),
+            "requests": self._n_requests,
             "rows_per_batch": (sum(self._batch_rows) / max(1, len(self._batch_rows))),
             "server_p50_ms": pick(0.50),
             "server_p99_ms": pick(0.99),
@@ -131,11 +136,11 @@ class DynamicBatcher:
                         self._cv.wait(timeout=remaining)
             batch, rows = [], 0
             while self._queue and rows + self._queue[0].n_rows <= self.max_batch_size:
-                req = self._queue.pop(0)
+                req = self._queue.popleft()
                 rows += req.n_rows
                 batch.append(req)
             if not batch and self._queue:  # single oversized request
-                batch.append(self._queue.pop(0))
+                batch.append(self._queue.popleft())
             return batch
 
     def _predict_batch(self, features_list: List[Any]):
@@ -195,6 +200,7 @@ class DynamicBatcher:
                 self._n_batches += 1
                 self._batch_rows.append(sum(r.n_rows for r in batch))
                 for req, res in zip(batch, results):
+                    self._n_requests += 1
                     self._done_latency_ms.append((now - req.t_submit) * 1000.0)
                     req.loop.call_soon_threadsafe(req.future.set_result, _jsonable(res))
             except Exception as exc:
