@@ -19,6 +19,7 @@ ext = CUDAExtension(
     sources=[
         "unionml_amd/ops/hip/tabular_ops.cpp",
         "unionml_amd/ops/hip/tabular_kernels.hip",
+        "unionml_amd/ops/hip/tabular_gen.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

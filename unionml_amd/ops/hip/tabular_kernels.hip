@@ -168,15 +168,18 @@ standardize_fit_kernel(const float* __restrict__ X, long long N, int D,
   }
 }
 
+// Dout >= D: output rows are Dout wide (extra columns untouched — the
+// generalized kernels stage features into a zero-padded [N][INP] image)
 extern "C" __global__ void __launch_bounds__(256)
 standardize_apply_kernel(const float* __restrict__ X, long long n_elems, int D,
+                         int Dout,
                          const float* __restrict__ mean,
                          const float* __restrict__ invstd,
                          u16* __restrict__ out) {
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n_elems;
        i += (long long)gridDim.x * blockDim.x) {
     const int col = (int)(i % D);
-    out[i] = f2bf((X[i] - mean[col]) * invstd[col]);
+    out[(i / D) * Dout + col] = f2bf((X[i] - mean[col]) * invstd[col]);
   }
 }
 
@@ -613,10 +616,12 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
     L.loss[0] = 0.f;
     // read the epoch source BEFORE any workgroup of this launch can have
     // advanced it (the writer only writes after every WG has published,
-    // i.e. after every WG has passed this point). Adam mode: the Adam
-    // step counter t_dev; reduce-only mode: the dedicated launch counter
-    // (t_dev is advanced by the separate adam_step kernel there).
-    lossu[2] = grads_out ? *counter : (unsigned)(*t_dev);
+    // i.e. after every WG has passed this point). BOTH modes use the
+    // dedicated launch counter, so mixing fused-Adam and reduce-only
+    // launches on one model keeps epochs unique and monotonic; t_dev is
+    // read separately for Adam bias correction only.
+    lossu[2] = *counter;
+    lossu[3] = (unsigned)(*t_dev);
   }
   zero_dl_pad(L);
 #ifndef PROBE_SKIP_LOADS
@@ -729,12 +734,13 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
 
 #ifdef PROBE_SKIP_REDUCE
   if (tid == 0 && blockIdx.x == 0) {
-    if (grads_out) *counter = epoch; else *t_dev = (int)(lossu[2] + 1u);
+    *counter = epoch;
+    if (!grads_out) *t_dev = (int)(lossu[3] + 1u);
   }
   return;
 #endif
   // ---- every WG reduces its own param stripe + applies Adam ----------------
-  const float t_new = (float)(lossu[2] + 1u);
+  const float t_new = (float)(lossu[3] + 1u);
   const float corr1 = fast_rcp(1.f - __powf(beta1, t_new));
   const float corr2 = fast_rcp(1.f - __powf(beta2, t_new));
   const int span = (NPARAM + 1 + n_wg - 1) / n_wg;
@@ -779,8 +785,8 @@ mlp_step_fused_kernel(const u16* __restrict__ Xbf, const int* __restrict__ y,
     if (wimg) wimg_write(wimg, i, wb);
   }
   if (tid == 0 && blockIdx.x == 0) {
-    if (grads_out) *counter = epoch;
-    else *t_dev = (int)t_new;
+    *counter = epoch;
+    if (!grads_out) *t_dev = (int)t_new;
   }
 }
 
@@ -1112,14 +1118,14 @@ void launch_standardize_fit(const float* X, long long N, int D, float* mean,
                      X, N, D, mean, invstd, eps);
 }
 
-void launch_standardize_apply(const float* X, long long N, int D,
+void launch_standardize_apply(const float* X, long long N, int D, int Dout,
                               const float* mean, const float* invstd,
                               unsigned short* out, hipStream_t stream) {
   const long long n = N * D;
   int blocks = (int)((n + 255) / 256);
   if (blocks > 2048) blocks = 2048;
   hipLaunchKernelGGL(standardize_apply_kernel, dim3(blocks), dim3(256), 0, stream,
-                     X, n, D, mean, invstd, out);
+                     X, n, D, Dout, mean, invstd, out);
 }
 
 void launch_mlp_step(const unsigned short* Xbf, const int* y, int B,

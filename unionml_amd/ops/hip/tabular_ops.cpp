@@ -8,8 +8,20 @@
 extern "C" {
 void launch_standardize_fit(const float*, long long, int, float*, float*, float,
                             double*, hipStream_t);
-void launch_standardize_apply(const float*, long long, int, const float*,
+void launch_standardize_apply(const float*, long long, int, int, const float*,
                               const float*, unsigned short*, hipStream_t);
+// generalized-geometry kernels (tabular_gen.hip)
+int gen_rt_for_hid(int);
+int launch_mlp_step_gen(const unsigned short*, const int*, int, int, int, int,
+                        unsigned short*, float*, unsigned short*, float*, float*,
+                        int*, float*, int, int, unsigned*, float*, float, float,
+                        float, float, float, float*, hipStream_t);
+int launch_mlp_predict_gen(const float*, int, int, int, int, int, const float*,
+                           const float*, const unsigned short*, const float*,
+                           int*, float*, hipStream_t);
+void launch_adam_step_gen(float*, unsigned short*, const float*, float*, float*,
+                          int*, int, int, int, float, float, float, float,
+                          unsigned short*, hipStream_t);
 void launch_mlp_step(const unsigned short*, const int*, int, const unsigned short*,
                      const unsigned short*, const float*, float*, float,
                      hipStream_t);
@@ -72,10 +84,13 @@ void standardize_apply(torch::Tensor X, torch::Tensor mean, torch::Tensor invstd
                        torch::Tensor out_bf16) {
   check(X, torch::kFloat32, "X");
   check(out_bf16, torch::kBFloat16, "out");
-  TORCH_CHECK(out_bf16.numel() == X.numel(), "out size mismatch");
+  TORCH_CHECK(out_bf16.dim() == 2 && out_bf16.size(0) == X.size(0) &&
+                  out_bf16.size(1) >= X.size(1),
+              "out must be [N][Dout] with Dout >= D");
   launch_standardize_apply(X.data_ptr<float>(), X.size(0), (int)X.size(1),
-                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                           bf16_mut_ptr(out_bf16), current_stream());
+                           (int)out_bf16.size(1), mean.data_ptr<float>(),
+                           invstd.data_ptr<float>(), bf16_mut_ptr(out_bf16),
+                           current_stream());
 }
 
 void mlp_step(torch::Tensor Xbf, torch::Tensor y, torch::Tensor W1bf,
@@ -198,6 +213,98 @@ void adam_step(torch::Tensor master, torch::Tensor bfmirror, torch::Tensor grads
                    (float)eps, wimg_ptr, current_stream());
 }
 
+// ---------------------------------------------------------------------------
+// generalized-geometry (any in_features/hidden/classes) entry points
+// ---------------------------------------------------------------------------
+
+bool mlp_step_gen(torch::Tensor Xbf, torch::Tensor y, int64_t hid, int64_t cls,
+                  torch::Tensor wimg, torch::Tensor master, torch::Tensor bfmirror,
+                  torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
+                  torch::Tensor slabs, torch::Tensor counter,
+                  torch::Tensor loss_out, double invBtot, double lr,
+                  double beta1, double beta2, double eps,
+                  c10::optional<torch::Tensor> grads_out = c10::nullopt) {
+  check(Xbf, torch::kBFloat16, "Xbf");
+  check(y, torch::kInt32, "y");
+  check(wimg, torch::kBFloat16, "wimg");
+  check(master, torch::kFloat32, "master");
+  check(bfmirror, torch::kBFloat16, "bfmirror");
+  check(slabs, torch::kFloat32, "slabs");
+  check(counter, torch::kUInt32, "counter");
+  check(loss_out, torch::kFloat32, "loss_out");
+  const int inp = (int)Xbf.size(1);
+  const int nparam = inp * (int)hid + (int)hid + (int)hid * 16 + 16;
+  TORCH_CHECK(inp % 32 == 0, "staged input width must be a multiple of 32");
+  TORCH_CHECK(master.numel() >= nparam, "master too small for geometry");
+  TORCH_CHECK(wimg.numel() >= (int64_t)hid * inp + hid * 32 + 16 * hid,
+              "wimg too small for geometry");
+  TORCH_CHECK(slabs.dim() == 2 && slabs.size(1) >= nparam + 2,
+              "slabs must be [n][>= nparam+2]");
+  float* grads_ptr = nullptr;
+  if (grads_out.has_value()) {
+    check(*grads_out, torch::kFloat32, "grads_out");
+    TORCH_CHECK(grads_out->numel() >= nparam + 1, "grads_out too small");
+    grads_ptr = grads_out->data_ptr<float>();
+  }
+  const int rc = launch_mlp_step_gen(
+      bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0), inp, (int)hid,
+      (int)cls, bf16_mut_ptr(wimg), master.data_ptr<float>(),
+      bf16_mut_ptr(bfmirror), m.data_ptr<float>(), v.data_ptr<float>(),
+      t_dev.data_ptr<int>(), slabs.data_ptr<float>(), (int)slabs.size(1),
+      (int)slabs.size(0), (unsigned*)counter.data_ptr(),
+      loss_out.data_ptr<float>(), (float)invBtot, (float)lr, (float)beta1,
+      (float)beta2, (float)eps, grads_ptr, current_stream());
+  TORCH_CHECK(rc != -2, "mlp_step_gen: hipFuncSetAttribute(LDS) failed");
+  TORCH_CHECK(rc != -3, "mlp_step_gen: unsupported geometry (hid=", hid,
+              " inp=", inp, " cls=", cls, ")");
+  return rc == 0;  // false -> more WGs than slab rows, caller re-sizes
+}
+
+void mlp_predict_gen(torch::Tensor X, int64_t inp, int64_t hid, int64_t cls,
+                     torch::Tensor mean, torch::Tensor invstd, torch::Tensor wimg,
+                     torch::Tensor master, torch::Tensor preds,
+                     c10::optional<torch::Tensor> probs) {
+  check(X, torch::kFloat32, "X");
+  check(wimg, torch::kBFloat16, "wimg");
+  check(master, torch::kFloat32, "master");
+  check(preds, torch::kInt32, "preds");
+  TORCH_CHECK(mean.numel() == X.size(1) && invstd.numel() == X.size(1),
+              "mean/invstd must match raw feature width");
+  float* probs_ptr = nullptr;
+  if (probs.has_value()) {
+    check(*probs, torch::kFloat32, "probs");
+    probs_ptr = probs->data_ptr<float>();
+  }
+  const int rc = launch_mlp_predict_gen(
+      X.data_ptr<float>(), (int)X.size(0), (int)X.size(1), (int)inp, (int)hid,
+      (int)cls, mean.data_ptr<float>(), invstd.data_ptr<float>(), bf16_ptr(wimg),
+      master.data_ptr<float>(), preds.data_ptr<int>(), probs_ptr,
+      current_stream());
+  TORCH_CHECK(rc == 0, "mlp_predict_gen: unsupported geometry or LDS failure");
+}
+
+void adam_step_gen(torch::Tensor master, torch::Tensor bfmirror,
+                   torch::Tensor grads, torch::Tensor m, torch::Tensor v,
+                   torch::Tensor t_dev, int64_t inp, int64_t hid, double lr,
+                   double beta1, double beta2, double eps,
+                   c10::optional<torch::Tensor> wimg = c10::nullopt) {
+  check(master, torch::kFloat32, "master");
+  check(bfmirror, torch::kBFloat16, "bfmirror");
+  check(grads, torch::kFloat32, "grads");
+  const int nparam = (int)(inp * hid + hid + hid * 16 + 16);
+  TORCH_CHECK(master.numel() >= nparam, "master too small for geometry");
+  unsigned short* wimg_ptr = nullptr;
+  if (wimg.has_value()) {
+    check(*wimg, torch::kBFloat16, "wimg");
+    wimg_ptr = bf16_mut_ptr(*wimg);
+  }
+  launch_adam_step_gen(master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
+                       grads.data_ptr<float>(), m.data_ptr<float>(),
+                       v.data_ptr<float>(), t_dev.data_ptr<int>(), nparam,
+                       (int)inp, (int)hid, (float)lr, (float)beta1, (float)beta2,
+                       (float)eps, wimg_ptr, current_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("standardize_fit", &standardize_fit, "column mean/invstd (CDNA4)");
   m.def("standardize_apply", &standardize_apply, "(x-mean)*invstd -> bf16 (CDNA4)");
@@ -218,4 +325,25 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("master"), py::arg("bfmirror"), py::arg("grads"), py::arg("m"),
         py::arg("v"), py::arg("t_dev"), py::arg("lr"), py::arg("beta1"),
         py::arg("beta2"), py::arg("eps"), py::arg("wimg") = c10::nullopt);
+  m.def("gen_rt_for_hid", &gen_rt_for_hid,
+        "rows-per-workgroup for a supported hidden width (0 = unsupported)");
+  m.def("mlp_step_gen", &mlp_step_gen,
+        "generalized fused step: any (in,hid,cls) geometry; K-tiled MFMA "
+        "fwd+bwd + slab reduction + Adam in one launch (grads_out: reduce-only)",
+        py::arg("Xbf"), py::arg("y"), py::arg("hid"), py::arg("cls"),
+        py::arg("wimg"), py::arg("master"), py::arg("bfmirror"), py::arg("m"),
+        py::arg("v"), py::arg("t_dev"), py::arg("slabs"), py::arg("counter"),
+        py::arg("loss_out"), py::arg("invBtot"), py::arg("lr"), py::arg("beta1"),
+        py::arg("beta2"), py::arg("eps"), py::arg("grads_out") = c10::nullopt);
+  m.def("mlp_predict_gen", &mlp_predict_gen,
+        "generalized fused standardize+fwd+argmax",
+        py::arg("X"), py::arg("inp"), py::arg("hid"), py::arg("cls"),
+        py::arg("mean"), py::arg("invstd"), py::arg("wimg"), py::arg("master"),
+        py::arg("preds"), py::arg("probs") = c10::nullopt);
+  m.def("adam_step_gen", &adam_step_gen,
+        "generalized fused Adam (runtime param count; DP path)",
+        py::arg("master"), py::arg("bfmirror"), py::arg("grads"), py::arg("m"),
+        py::arg("v"), py::arg("t_dev"), py::arg("inp"), py::arg("hid"),
+        py::arg("lr"), py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+        py::arg("wimg") = c10::nullopt);
 }

@@ -1,17 +1,30 @@
-"""TabularMLP — the framework's default tabular hot path.
+"""TabularMLP — the framework's default tabular hot path, for ANY
+(in_features, hidden, classes) geometry.
 
-A 64 -> 32(relu) -> 10(softmax) classifier whose training step
-(standardize, fused Linear+Softmax forward, cross-entropy backward,
-fused Adam) runs as hand-written CDNA4 HIP kernels on MFMA
-(unionml_amd/ops/hip/tabular_kernels.hip). The per-epoch minibatch loop
-is captured ONCE into a hipGraph and replayed per epoch — zero launch
-overhead in steady state. On CPU the same class runs the pure-torch
-reference path (unionml_amd/ops/reference.py): one user-visible code
-path, two substrates.
+A one-hidden-layer classifier (Linear -> ReLU -> Linear -> Softmax)
+whose training step (standardize, fused forward, cross-entropy
+backward, fused Adam) runs as hand-written CDNA4 HIP kernels on MFMA:
+
+* the hand-tuned specialized kernels for the 64->32->10 digits shape
+  (unionml_amd/ops/hip/tabular_kernels.hip — the headline benchmark
+  path, with the persistent and packed-weight-image engines), and
+* the generalized templated kernels for every other shape
+  (unionml_amd/ops/hip/tabular_gen.hip — input width streamed through
+  LDS in k-tiles, hidden width dispatched over compiled instantiations
+  {32,64,128,256}, odd sizes zero-padded exactly).
+
+The per-epoch minibatch loop is captured ONCE into a hipGraph and
+replayed per epoch — zero launch overhead in steady state. On CPU the
+same class runs the pure-torch reference path
+(unionml_amd/ops/reference.py): one user-visible code path, two
+substrates.
 
 Data parallel: under an active torch.distributed process group each
-rank trains its row shard and the flat 2.6k-float gradient buffer is
-all-reduced on RCCL between the step and Adam kernels.
+rank trains its row shard and the flat gradient buffer is all-reduced
+on RCCL between the reduce-only step and the Adam kernel.
+
+Reference behavior mirrored (no code ported — the reference is pure
+Python): tests/integration/pytorch_app/quickstart.py:14-70.
 """
 
 import math
@@ -23,6 +36,7 @@ from unionml_amd._logging import logger
 from unionml_amd.ops import hip_available, hip_ext
 from unionml_amd.ops import reference as ref
 from unionml_amd.ops.reference import CLS, CPAD, HID, IN, NPARAM, OFF_B1, OFF_B2, OFF_W1, OFF_W2
+from unionml_amd.ops.reference import Geometry
 
 ADAM_BETA1, ADAM_BETA2, ADAM_EPS = 0.9, 0.999, 1e-8
 
@@ -30,70 +44,109 @@ ADAM_BETA1, ADAM_BETA2, ADAM_EPS = 0.9, 0.999, 1e-8
 class TabularMLP:
     """Device-resident parameter/optimizer state + fused train/predict."""
 
-    def __init__(self, device: Optional[str] = None, seed: int = 0):
+    def __init__(
+        self,
+        in_features: int = 64,
+        hidden: int = 32,
+        classes: int = 10,
+        device: Optional[str] = None,
+        seed: int = 0,
+    ):
+        self.g = Geometry(in_features, hidden, classes)
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         self.device = torch.device(device)
         self.use_hip = self.device.type == "cuda"
         if self.use_hip:
             hip_ext(required=True)  # loud failure if the gfx950 ext is missing
+        # hand-tuned specialized kernels for the digits shape; generalized
+        # templated kernels for everything else
+        self.use_spec = self.g.is_specialized
 
-        gen = torch.Generator().manual_seed(seed)
-        master = torch.zeros(NPARAM, dtype=torch.float32)
-        W1, b1, W2, b2 = ref.unpack_master(master)
-        W1.copy_(torch.randn(IN, HID, generator=gen) * math.sqrt(2.0 / IN))
-        W2[:, :CLS] = torch.randn(HID, CLS, generator=gen) * math.sqrt(2.0 / HID)
+        g = self.g
+        gen_t = torch.Generator().manual_seed(seed)
+        master = torch.zeros(g.nparam, dtype=torch.float32)
+        W1, b1, W2, b2 = ref.unpack_master_g(g, master)
+        W1[: g.in_features, : g.hidden] = (
+            torch.randn(g.in_features, g.hidden, generator=gen_t)
+            * math.sqrt(2.0 / g.in_features)
+        )
+        W2[: g.hidden, : g.classes] = (
+            torch.randn(g.hidden, g.classes, generator=gen_t) * math.sqrt(2.0 / g.hidden)
+        )
 
         self.master = master.to(self.device)
         self.bfmirror = self.master.bfloat16()
         self.m = torch.zeros_like(self.master)
         self.v = torch.zeros_like(self.master)
         self.t_dev = torch.zeros(1, dtype=torch.int32, device=self.device)
-        self.grads = torch.zeros(NPARAM + 1, dtype=torch.float32, device=self.device)
-        self.mean = torch.zeros(IN, dtype=torch.float32, device=self.device)
-        self.invstd = torch.ones(IN, dtype=torch.float32, device=self.device)
+        self.grads = torch.zeros(g.nparam + 1, dtype=torch.float32, device=self.device)
+        self.mean = torch.zeros(g.in_features, dtype=torch.float32, device=self.device)
+        self.invstd = torch.ones(g.in_features, dtype=torch.float32, device=self.device)
         self.slabs = None       # per-WG partial-grad slabs for the fused step
-        self.counter = None     # G16 ticket counter (self-resetting)
+        self.counter = None     # G16 epoch counter (shared by both step modes)
         self._graph = None
         self._graph_key = None
-        # packed weight images (the three LDS layouts back to back) so
-        # the training prologue is a straight vectorized copy; kept in
-        # sync by the Adam kernels (wimg=) and rebuilt on any host-side
-        # weight mutation
+        # packed weight images (kernel-layout staging) so the training
+        # prologue is a straight vectorized copy; kept in sync by the
+        # Adam kernels (wimg=) and rebuilt on any host-side weight
+        # mutation. Specialized and generalized kernels use different
+        # packings (LDS-stride padded vs dense row-major).
         self.wimg = (
-            torch.zeros(4224, dtype=torch.bfloat16, device=self.device)
+            torch.zeros(self._wimg_len(), dtype=torch.bfloat16, device=self.device)
             if self.use_hip
             else None
         )
         self._build_wimg()
 
-    _XS, _WS = 72, 40  # LDS row strides (tabular_kernels.hip)
+    _XS, _WS = 72, 40  # specialized-kernel LDS row strides (tabular_kernels.hip)
+
+    def _wimg_len(self) -> int:
+        return 4224 if self.use_spec else self.g.wimg_n
+
+    def _rows_per_wg(self) -> int:
+        if self.use_spec:
+            return 128
+        return {32: 128, 64: 128, 128: 64, 256: 32}[self.g.hid]
 
     def _build_wimg(self):
         if self.wimg is None:
             return
-        XS, WS = self._XS, self._WS
-        W1, _, W2, _ = ref.unpack_master(self.master.detach().cpu())
-        img = torch.zeros(4224, dtype=torch.bfloat16)
-        img[: HID * XS].view(HID, XS)[:, :IN] = W1.t().bfloat16()      # W1T[h][k]
-        img[HID * XS : HID * XS + HID * WS].view(HID, WS)[:, :CPAD] = W2.bfloat16()  # W2s[h][c]
-        img[HID * XS + HID * WS :].view(CPAD, WS)[:, :HID] = W2.t().bfloat16()       # W2T[c][h]
+        g = self.g
+        W1, _, W2, _ = ref.unpack_master_g(g, self.master.detach().cpu())
+        if self.use_spec:
+            XS, WS = self._XS, self._WS
+            img = torch.zeros(4224, dtype=torch.bfloat16)
+            img[: HID * XS].view(HID, XS)[:, :IN] = W1.t().bfloat16()      # W1T[h][k]
+            img[HID * XS : HID * XS + HID * WS].view(HID, WS)[:, :CPAD] = W2.bfloat16()
+            img[HID * XS + HID * WS :].view(CPAD, WS)[:, :HID] = W2.t().bfloat16()
+        else:
+            img = torch.zeros(g.wimg_n, dtype=torch.bfloat16)
+            o1 = g.hid * g.inp
+            o2 = o1 + g.hid * 32
+            img[:o1].view(g.hid, g.inp).copy_(W1.t().bfloat16())           # W1T[h][k]
+            img[o1:o2].view(g.hid, 32)[:, :CPAD] = W2.bfloat16()           # W2s K-pad
+            img[o2:].view(CPAD, g.hid).copy_(W2.t().bfloat16())            # W2T[c][h]
         self.wimg.copy_(img.to(self.device))
 
     def _ensure_slabs(self, n_wg: int):
         if self.slabs is None or self.slabs.shape[0] < n_wg:
-            self.slabs = torch.zeros(n_wg, 2624, dtype=torch.float32, device=self.device)
+            self.slabs = torch.zeros(
+                n_wg, self.g.slab_stride, dtype=torch.float32, device=self.device
+            )
             self.counter = torch.zeros(1, dtype=torch.uint32, device=self.device)
 
     # -- views ----------------------------------------------------------------
 
     @property
     def W1bf(self) -> torch.Tensor:
-        return self.bfmirror[OFF_W1 : OFF_W1 + IN * HID].view(IN, HID)
+        g = self.g
+        return self.bfmirror[g.off_w1 : g.off_w1 + g.inp * g.hid].view(g.inp, g.hid)
 
     @property
     def W2bf(self) -> torch.Tensor:
-        return self.bfmirror[OFF_W2 : OFF_W2 + HID * CPAD].view(HID, CPAD)
+        g = self.g
+        return self.bfmirror[g.off_w2 : g.off_w2 + g.hid * g.cpad].view(g.hid, g.cpad)
 
     # -- standardizer ----------------------------------------------------------
 
@@ -107,74 +160,127 @@ class TabularMLP:
             self.invstd.copy_(invstd)
 
     def stage(self, X: torch.Tensor) -> torch.Tensor:
-        """Standardize fp32 features into a device-resident bf16 matrix."""
+        """Standardize fp32 features into a device-resident bf16 matrix,
+        zero-padded to the kernel's input width."""
+        g = self.g
         X = X.to(self.device, torch.float32).contiguous()
         if self.use_hip:
-            out = torch.empty(X.shape, dtype=torch.bfloat16, device=self.device)
+            if g.inp == g.in_features:
+                out = torch.empty(X.shape[0], g.inp, dtype=torch.bfloat16, device=self.device)
+            else:
+                out = torch.zeros(X.shape[0], g.inp, dtype=torch.bfloat16, device=self.device)
             hip_ext().standardize_apply(X, self.mean, self.invstd, out)
             return out
-        return ref.standardize_apply(X, self.mean, self.invstd)
+        out_raw = ref.standardize_apply(X, self.mean, self.invstd)
+        if g.inp == g.in_features:
+            return out_raw
+        out = torch.zeros(X.shape[0], g.inp, dtype=torch.bfloat16)
+        out[:, : g.in_features] = out_raw
+        return out
 
     # -- training --------------------------------------------------------------
 
-    def _step(self, Xbf: torch.Tensor, y: torch.Tensor, invBtot: float, lr: float,
-              allreduce: bool):
-        if self.use_hip:
-            # fused fwd/bwd + in-kernel slab reduction writing the summed
-            # grads (reduce-only mode: no global atomics, no zeroing pass)
-            self._ensure_slabs((Xbf.shape[0] + 127) // 128)
-            loss_out = self.grads[NPARAM : NPARAM + 1]
-            ok = hip_ext().mlp_step_fused(
+    def _step_reduce(self, Xbf: torch.Tensor, y: torch.Tensor, invBtot: float, lr: float):
+        """Fused fwd/bwd + in-kernel slab reduction writing the summed
+        grads into self.grads (reduce-only mode: the DP pre-collective
+        kernel — no Adam)."""
+        g = self.g
+        self._ensure_slabs((Xbf.shape[0] + self._rows_per_wg() - 1) // self._rows_per_wg())
+        loss_out = self.grads[g.nparam : g.nparam + 1]
+        ext = hip_ext()
+        if self.use_spec:
+            ok = ext.mlp_step_fused(
                 Xbf, y, self.W1bf, self.W2bf, self.master, self.bfmirror,
                 self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
                 invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
                 grads_out=self.grads, wimg=self.wimg,
             )
-            assert ok, "mlp_step_fused slab capacity exceeded"
+        else:
+            ok = ext.mlp_step_gen(
+                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.bfmirror,
+                self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
+                invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                grads_out=self.grads,
+            )
+        assert ok, "fused step slab capacity exceeded"
+
+    def _adam(self, lr: float):
+        g = self.g
+        ext = hip_ext()
+        if self.use_spec:
+            ext.adam_step(
+                self.master, self.bfmirror, self.grads, self.m, self.v, self.t_dev,
+                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
+            )
+        else:
+            ext.adam_step_gen(
+                self.master, self.bfmirror, self.grads, self.m, self.v, self.t_dev,
+                g.inp, g.hid, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
+            )
+
+    def _step(self, Xbf: torch.Tensor, y: torch.Tensor, invBtot: float, lr: float,
+              allreduce: bool):
+        g = self.g
+        if self.use_hip:
+            self._step_reduce(Xbf, y, invBtot, lr)
         else:
             self.grads.zero_()
-            ref.mlp_step(Xbf, y, self.W1bf, self.W2bf, self.master, self.grads, invBtot)
+            ref.mlp_step_g(g, Xbf, y, self.W1bf, self.W2bf, self.master, self.grads, invBtot)
         if allreduce:
             import torch.distributed as dist
 
             dist.all_reduce(self.grads)
         if self.use_hip:
-            hip_ext().adam_step(
-                self.master, self.bfmirror, self.grads, self.m, self.v, self.t_dev,
-                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
-            )
+            self._adam(lr)
         else:
             self.t_dev += 1
-            ref.adam_step(
-                self.master, self.bfmirror, self.grads, self.m, self.v,
+            ref.adam_step_g(
+                g, self.master, self.bfmirror, self.grads, self.m, self.v,
                 int(self.t_dev.item()), lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
             )
-        return self.grads[NPARAM]
+        return self.grads[g.nparam]
+
+    def _fused_adam_step(self, Xbf, y, invBtot, lr, loss_out):
+        """One fully-fused optimizer step (fwd+bwd+reduce+Adam in one
+        launch for the specialized shape; one launch + image update for
+        generalized shapes)."""
+        g = self.g
+        ext = hip_ext()
+        if self.use_spec:
+            ok = ext.mlp_step_fused(
+                Xbf, y, self.W1bf, self.W2bf, self.master, self.bfmirror,
+                self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
+                invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
+            )
+        else:
+            ok = ext.mlp_step_gen(
+                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.bfmirror,
+                self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
+                invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+            )
+        assert ok, "fused step slab capacity exceeded"
 
     def _train_epochs_fused(self, Xbf, y, batches, *, epochs, lr, use_graph) -> float:
-        ext = hip_ext()
-        self._ensure_slabs(max((bs + 127) // 128 for _, bs in batches))
-        loss_out = self.grads[NPARAM : NPARAM + 1]
+        g = self.g
+        rpw = self._rows_per_wg()
+        self._ensure_slabs(max((bs + rpw - 1) // rpw for _, bs in batches))
+        loss_out = self.grads[g.nparam : g.nparam + 1]
 
         def run_epoch():
             for off, bs in batches:
-                ok = ext.mlp_step_fused(
-                    Xbf[off : off + bs], y[off : off + bs], self.W1bf, self.W2bf,
-                    self.master, self.bfmirror, self.m, self.v, self.t_dev,
-                    self.slabs, self.counter, loss_out, 1.0 / bs,
-                    lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
+                self._fused_adam_step(
+                    Xbf[off : off + bs], y[off : off + bs], 1.0 / bs, lr, loss_out
                 )
-                assert ok, "mlp_step_fused slab capacity exceeded"
 
         run_epoch()   # warmup epoch (eager)
         remaining = epochs - 1
         key = ("fused", len(batches), lr, Xbf.data_ptr(), y.data_ptr())
         if use_graph and self._graph_key != key:
             try:
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                gph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(gph):
                     run_epoch()
-                self._graph, self._graph_key = g, key
+                self._graph, self._graph_key = gph, key
             except RuntimeError as exc:
                 logger.warning("hipGraph capture failed (%s); eager stepping", exc)
                 self._graph, self._graph_key = None, None
@@ -204,6 +310,7 @@ class TabularMLP:
         Returns the last step's loss. ``world_size > 1`` means this rank
         holds a shard and gradients all-reduce over RCCL each step.
         """
+        g = self.g
         y = y.to(self.device, torch.int32).contiguous()
         n = Xbf.shape[0]
         batches = [
@@ -215,8 +322,13 @@ class TabularMLP:
         # cross-WG slab reduction + Adam in ONE launch), with the epoch's
         # minibatch loop captured into a hipGraph.
         if self.use_hip and not allreduce:
-            if engine == "persistent" and batch_size % 128 == 0 and n % batch_size == 0:
-                loss_out = self.grads[NPARAM : NPARAM + 1]
+            if (
+                self.use_spec
+                and engine == "persistent"
+                and batch_size % 128 == 0
+                and n % batch_size == 0
+            ):
+                loss_out = self.grads[g.nparam : g.nparam + 1]
                 n_steps = epochs * (n // batch_size)
                 ok = hip_ext().mlp_train_steps(
                     Xbf, y, batch_size, n_steps, self.master, self.bfmirror,
@@ -261,10 +373,10 @@ class TabularMLP:
         key = (n, batch_size, world_size, lr, Xbf.data_ptr(), y.data_ptr())
         if self._graph_key != key:
             try:
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
+                gph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(gph):
                     run_epoch()
-                self._graph, self._graph_key = g, key
+                self._graph, self._graph_key = gph, key
             except RuntimeError as exc:  # e.g. RCCL capture unsupported
                 logger.warning("hipGraph capture failed (%s); eager stepping", exc)
                 self._graph, self._graph_key = None, None
@@ -275,25 +387,33 @@ class TabularMLP:
             for _ in range(remaining):
                 run_epoch()
         torch.cuda.synchronize(self.device)
-        return float(self.grads[NPARAM].item())
+        return float(self.grads[g.nparam].item())
 
     # -- inference -------------------------------------------------------------
 
     def predict(self, X: torch.Tensor, return_probs: bool = False):
+        g = self.g
         X = X.to(self.device, torch.float32).contiguous()
         if self.use_hip:
             preds = torch.empty(X.shape[0], dtype=torch.int32, device=self.device)
             probs = (
-                torch.empty(X.shape[0], CLS, dtype=torch.float32, device=self.device)
+                torch.empty(X.shape[0], g.classes, dtype=torch.float32, device=self.device)
                 if return_probs
                 else None
             )
-            hip_ext().mlp_predict(
-                X, self.mean, self.invstd, self.W1bf, self.W2bf, self.master, preds, probs
-            )
+            if self.use_spec:
+                hip_ext().mlp_predict(
+                    X, self.mean, self.invstd, self.W1bf, self.W2bf, self.master,
+                    preds, probs,
+                )
+            else:
+                hip_ext().mlp_predict_gen(
+                    X, g.inp, g.hid, g.classes, self.mean, self.invstd, self.wimg,
+                    self.master, preds, probs,
+                )
             return (preds, probs) if return_probs else preds
-        return ref.mlp_predict(
-            X, self.mean, self.invstd, self.W1bf, self.W2bf, self.master, return_probs
+        return ref.mlp_predict_g(
+            g, X, self.mean, self.invstd, self.W1bf, self.W2bf, self.master, return_probs
         )
 
     # -- persistence -----------------------------------------------------------
@@ -317,6 +437,8 @@ class TabularMLP:
         device = torch.device(state.pop("device"))
         if device.type == "cuda" and not torch.cuda.is_available():
             device = torch.device("cpu")
+        state.setdefault("g", Geometry(64, 32, 10))       # pre-geometry pickles
+        state.setdefault("use_spec", state["g"].is_specialized)
         self.__dict__.update(state)
         self.device = device
         self.use_hip = device.type == "cuda"
@@ -326,27 +448,36 @@ class TabularMLP:
             if torch.is_tensor(v):
                 setattr(self, k, v.to(device))
         if self.use_hip and self.wimg is None:
-            self.wimg = torch.zeros(4224, dtype=torch.bfloat16, device=device)
+            self.wimg = torch.zeros(self._wimg_len(), dtype=torch.bfloat16, device=device)
         self._build_wimg()
 
     def state_dict(self) -> Dict[str, torch.Tensor]:
-        W1, b1, W2, b2 = ref.unpack_master(self.master.cpu())
+        """Logical (unpadded) weights + standardizer state."""
+        g = self.g
+        W1, b1, W2, b2 = ref.unpack_master_g(g, self.master.cpu())
         return {
-            "W1": W1.clone(),
-            "b1": b1.clone(),
-            "W2": W2[:, :CLS].clone(),
-            "b2": b2[:CLS].clone(),
+            "W1": W1[: g.in_features, : g.hidden].clone(),
+            "b1": b1[: g.hidden].clone(),
+            "W2": W2[: g.hidden, : g.classes].clone(),
+            "b2": b2[: g.classes].clone(),
             "mean": self.mean.cpu().clone(),
             "invstd": self.invstd.cpu().clone(),
+            "geometry": torch.tensor([g.in_features, g.hidden, g.classes]),
         }
 
     def load_state_dict(self, state: Dict[str, torch.Tensor]):
-        master = torch.zeros(NPARAM, dtype=torch.float32)
-        W1, b1, W2, b2 = ref.unpack_master(master)
-        W1.copy_(state["W1"])
-        b1.copy_(state["b1"])
-        W2[:, :CLS] = state["W2"]
-        b2[:CLS] = state["b2"]
+        g = self.g
+        if "geometry" in state:
+            want = tuple(int(x) for x in state["geometry"])
+            have = (g.in_features, g.hidden, g.classes)
+            if want != have:
+                raise ValueError(f"state geometry {want} != model geometry {have}")
+        master = torch.zeros(g.nparam, dtype=torch.float32)
+        W1, b1, W2, b2 = ref.unpack_master_g(g, master)
+        W1[: g.in_features, : g.hidden] = state["W1"]
+        b1[: g.hidden] = state["b1"]
+        W2[: g.hidden, : g.classes] = state["W2"]
+        b2[: g.classes] = state["b2"]
         self.master.copy_(master.to(self.device))
         self.bfmirror.copy_(self.master.bfloat16())
         self.mean.copy_(state["mean"].to(self.device))
