@@ -30,7 +30,7 @@ class TabularGraphRunner:
         self._buckets: Dict[int, tuple] = {}
         if self.use_graphs:
             self._pinned = torch.empty(
-                max_batch_size, 64, dtype=torch.float32, pin_memory=True
+                max_batch_size, mlp.g.in_features, dtype=torch.float32, pin_memory=True
             )
             if precapture:
                 # capture every bucket up front so a first request of a
@@ -47,21 +47,28 @@ class TabularGraphRunner:
 
         entry = self._buckets.get(b)
         if entry is None:
-            x = torch.zeros(b, 64, dtype=torch.float32, device=self.mlp.device)
-            preds = torch.zeros(b, dtype=torch.int32, device=self.mlp.device)
+            mlp = self.mlp
+            x = torch.zeros(b, mlp.g.in_features, dtype=torch.float32, device=mlp.device)
+            preds = torch.zeros(b, dtype=torch.int32, device=mlp.device)
             ext = hip_ext(required=True)
-            # warmup then capture
-            ext.mlp_predict(
-                x, self.mlp.mean, self.mlp.invstd, self.mlp.W1bf, self.mlp.W2bf,
-                self.mlp.master, preds, None,
-            )
-            torch.cuda.synchronize(self.mlp.device)
+
+            def fwd():
+                if mlp.use_spec:
+                    ext.mlp_predict(
+                        x, mlp.mean, mlp.invstd, mlp.W1bf, mlp.W2bf,
+                        mlp.master, preds, None,
+                    )
+                else:
+                    ext.mlp_predict_gen(
+                        x, mlp.g.inp, mlp.g.hid, mlp.g.classes, mlp.mean,
+                        mlp.invstd, mlp.wimg, mlp.master, preds, None,
+                    )
+
+            fwd()  # warmup then capture
+            torch.cuda.synchronize(mlp.device)
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                ext.mlp_predict(
-                    x, self.mlp.mean, self.mlp.invstd, self.mlp.W1bf, self.mlp.W2bf,
-                    self.mlp.master, preds, None,
-                )
+                fwd()
             entry = (g, x, preds)
             self._buckets[b] = entry
             logger.info("captured inference hipGraph for bucket %d", b)
