@@ -1,14 +1,17 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: digits-MLP training throughput (train samples/sec).
+"""Flagship benchmark: digits-MLP training throughput (train samples/sec)
+plus an optional serving leg (/predict p50).
 
 BASELINE.json metric: "train samples/sec + /predict p50 latency, MLP
-digits-shape, 1/2/4/8 MI355X". This measures the training leg on
-synthetic digits-shaped data (64 features, 10 classes, random-init
-weights — no network for datasets) with the CDNA4 fused hot path:
-one optimizer step = the fully-fused fwd/bwd/reduce/Adam MFMA kernel
-(or, under DP, fused fwd/bwd kernel + RCCL gradient all-reduce + fused
-Adam), with 64 optimizer steps captured per hipGraph so replay launch
-overhead is amortized (--graph-steps).
+digits-shape, 1/2/4/8 MI355X". The default mode measures the training
+leg on synthetic digits-shaped data (64 features, 10 classes,
+random-init weights — no network for datasets) with the CDNA4 fused hot
+path: one optimizer step = the fully-fused fwd/bwd/reduce/Adam MFMA
+kernel (or, under DP, fused fwd/bwd kernel + RCCL gradient all-reduce +
+fused Adam), with up to --graph-steps optimizer steps captured per
+hipGraph. ``--mode serve`` measures the /predict leg instead: real HTTP
+requests against the FastAPI serving app (dynamic batcher + bucketed
+hipGraph replay), reporting client-side p50 latency.
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W              # single GPU
@@ -17,6 +20,17 @@ Launch (driver contract):
 
 Weak scaling: per-GPU batch is fixed (default 2048); reported value is
 the WHOLE-JOB samples/sec aggregated over all ranks.
+
+Robustness rules (first-contact world>1 safety):
+  * hipGraph capture is VALIDATED before the timed loop: one step is
+    captured, replayed, and checksummed against the same step run
+    eagerly from identical state; mismatch or capture failure falls
+    back to eager — collectively across ranks (all-reduce MIN on the
+    verdict) so no rank replays graphs alone.
+  * engine fallbacks are per-engine, not per-process: unmet persistent-
+    kernel constraints drop to the fused engine instead of dying.
+  * the reported engine label reflects what actually ran in the timed
+    region (graph_replays_used is in the config blob).
 """
 
 import argparse
@@ -50,10 +64,29 @@ def parse_args():
         default="auto",
         help="auto: fused single-kernel step (1 GPU) / 3-kernel+RCCL (DP)",
     )
+    p.add_argument(
+        "--mode",
+        choices=["train", "serve"],
+        default="train",
+        help="train: optimizer-step throughput (driver default); "
+        "serve: /predict HTTP p50 over the FastAPI app",
+    )
+    p.add_argument("--serve-clients", type=int, default=1,
+                   help="concurrent HTTP clients in serve mode")
     return p.parse_args()
 
 
-def _result(args, n_gpus, B, elapsed, engine, loss):
+def _result(args, n_gpus, B, elapsed, engine, loss, replays=None):
+    cfg = {
+        "model": "digits_mlp_64x32x10",
+        "global_batch": B * n_gpus,
+        "seq_len": None,
+        "parallelism": f"dp{n_gpus}",
+        "engine": engine,
+        "final_loss": loss,
+    }
+    if replays is not None:
+        cfg["graph_replays_used"] = replays
     return {
         "metric": "train_samples_per_sec",
         "value": args.steps * B * n_gpus / elapsed,
@@ -67,19 +100,190 @@ def _result(args, n_gpus, B, elapsed, engine, loss):
         "vs_baseline": None,
         "dtype": "bf16",
         "data": "synthetic",
-        "config": {
-            "model": "digits_mlp_64x32x10",
-            "global_batch": B * n_gpus,
-            "seq_len": None,
-            "parallelism": f"dp{n_gpus}",
-            "engine": engine,
-            "final_loss": loss,
-        },
+        "config": cfg,
     }
+
+
+# ---------------------------------------------------------------------------
+# serve mode: boot the real FastAPI app + uvicorn, measure /predict p50
+# ---------------------------------------------------------------------------
+
+def run_serve_mode(args):
+    import socket
+    import subprocess
+    import urllib.request
+
+    steps = args.steps if args.steps else 2000
+    warmup = args.warmup if args.warmup else 200
+    port = 18321
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+
+    # train a quick artifact for the server to load (synthetic digits)
+    import tempfile
+
+    from unionml_amd.models.mlp import model as train_model
+
+    train_model.artifact = None
+    train_model.train(trainer_kwargs={"epochs": 5, "lr": 3e-3}, synthetic=True, n=4096)
+    artifact_path = os.path.join(tempfile.mkdtemp(prefix="unionml_bench_"), "model.pt")
+    train_model.save(artifact_path)
+
+    env = dict(os.environ)
+    env["UNIONML_MODEL_PATH"] = artifact_path
+    server = subprocess.Popen(
+        [sys.executable, "-m", "uvicorn", "--host", "127.0.0.1", "--port", str(port),
+         "--log-level", "warning", "unionml_amd.models.mlp_serve:app"],
+        env=env,
+    )
+    try:
+        # wait for readiness
+        rng_rows = [{f"p{i}": float((i * 7) % 16) for i in range(64)}]
+        body = json.dumps({"features": rng_rows}).encode()
+        deadline = time.time() + 120
+        while True:
+            try:
+                urllib.request.urlopen(f"http://127.0.0.1:{port}/health", timeout=2)
+                break
+            except Exception:
+                if time.time() > deadline:
+                    raise RuntimeError("serve-mode server never became healthy")
+                time.sleep(0.5)
+
+        def one_request():
+            req = urllib.request.Request(
+                f"http://127.0.0.1:{port}/predict", data=body,
+                headers={"Content-Type": "application/json"},
+            )
+            t0 = time.perf_counter()
+            with urllib.request.urlopen(req, timeout=10) as resp:
+                resp.read()
+            return (time.perf_counter() - t0) * 1000.0
+
+        for _ in range(warmup):
+            one_request()
+        lats = []
+        if args.serve_clients <= 1:
+            t_start = time.perf_counter()
+            for _ in range(steps):
+                lats.append(one_request())
+            elapsed = time.perf_counter() - t_start
+        else:
+            import concurrent.futures as cf
+
+            t_start = time.perf_counter()
+            with cf.ThreadPoolExecutor(max_workers=args.serve_clients) as pool:
+                lats = list(pool.map(lambda _: one_request(), range(steps)))
+            elapsed = time.perf_counter() - t_start
+        lats.sort()
+        p = lambda q: lats[min(len(lats) - 1, int(q * len(lats)))]  # noqa: E731
+        print(json.dumps({
+            "metric": "predict_p50_ms",
+            "value": p(0.50),
+            "unit": "ms",
+            "n_gpus": 1,
+            "steps": steps,
+            "warmup": warmup,
+            "ms_per_step": elapsed / steps * 1000.0,
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "digits_mlp_64x32x10",
+                "global_batch": 1,
+                "seq_len": None,
+                "parallelism": "serve1",
+                "clients": args.serve_clients,
+                "p90_ms": p(0.90),
+                "p99_ms": p(0.99),
+                "rps": steps / elapsed,
+            },
+        }))
+    finally:
+        server.terminate()
+        server.wait(timeout=10)
+
+
+# ---------------------------------------------------------------------------
+# train mode
+# ---------------------------------------------------------------------------
+
+def snapshot_state(clf):
+    keys = ["master", "bfmirror", "m", "v", "t_dev", "grads"]
+    s = {k: getattr(clf, k).clone() for k in keys}
+    for opt in ("counter", "wimg"):
+        t = getattr(clf, opt, None)
+        if t is not None:
+            s[opt] = t.clone()
+    return s
+
+
+def restore_state(clf, s):
+    for k, v in s.items():
+        getattr(clf, k).copy_(v)
+
+
+def checksum_state(clf):
+    return (
+        float(clf.master.double().sum().item()),
+        float(clf.master.double().abs().sum().item()),
+        int(clf.t_dev.item()),
+    )
+
+
+def validate_graph_capture(clf, eager_step, dist, device):
+    """Capture one optimizer step, replay it, and compare the resulting
+    state against the same step run eagerly from identical initial
+    state. Returns True only if EVERY rank validates (collective MIN),
+    so no rank trusts graphs alone. RCCL-inside-hipGraph is exactly the
+    kind of thing that breaks at world=8 first — never enter the timed
+    loop on an unvalidated capture."""
+    s0 = snapshot_state(clf)
+    eager_step(0)
+    torch.cuda.synchronize()
+    want = checksum_state(clf)
+    restore_state(clf, s0)
+    torch.cuda.synchronize()
+
+    ok = True
+    try:
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            eager_step(0)
+        g.replay()
+        torch.cuda.synchronize()
+        got = checksum_state(clf)
+        ok = (
+            got[2] == want[2]
+            and abs(got[0] - want[0]) <= 1e-6 * max(1.0, abs(want[0]))
+            and abs(got[1] - want[1]) <= 1e-6 * max(1.0, want[1])
+        )
+        if not ok:
+            print(f"[bench] graph validation mismatch: eager={want} replay={got}",
+                  file=sys.stderr)
+        del g
+    except RuntimeError as exc:
+        print(f"[bench] graph capture validation failed ({exc})", file=sys.stderr)
+        ok = False
+    restore_state(clf, s0)
+    torch.cuda.synchronize()
+
+    if dist is not None:
+        flag = torch.tensor([1.0 if ok else 0.0], device=device)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+        ok = bool(flag.item() > 0.5)
+    return ok
 
 
 def main():
     args = parse_args()
+    if args.mode == "serve":
+        run_serve_mode(args)
+        return
+
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
@@ -131,6 +335,14 @@ def main():
     engine = args.engine
     if engine == "auto":
         engine = "fused" if (use_gpu and world == 1) else "stepwise"
+    # per-engine fallback: unmet persistent-kernel constraints drop to
+    # the fused engine instead of aborting the whole process
+    if engine == "persistent" and not (
+        use_gpu and world == 1 and B % 128 == 0 and (B * M) % B == 0
+    ):
+        print("[bench] persistent engine constraints unmet; falling back to fused",
+              file=sys.stderr)
+        engine = "fused" if (use_gpu and world == 1) else "stepwise"
 
     if use_gpu:
         from unionml_amd.ops import hip_ext
@@ -180,47 +392,62 @@ def main():
 
     # persistent single-workgroup engine: K steps in ONE kernel launch
     if use_gpu and world == 1 and engine == "persistent":
-        def run_steps(k):
-            ok = ext.mlp_train_steps(
+        def run_persistent(k):
+            return ext.mlp_train_steps(
                 Xbf, y, B, k, clf.master, clf.bfmirror, clf.m, clf.v,
                 clf.t_dev, loss_out, args.lr, 0.9, 0.999, 1e-8,
             )
-            assert ok, "mlp_train_steps constraints unmet"
 
-        run_steps(args.warmup)
-        torch.cuda.synchronize()
-        t0 = time.perf_counter()
-        run_steps(args.steps)
-        torch.cuda.synchronize()
-        elapsed = time.perf_counter() - t0
-        loss = float(loss_out.item())
-        assert loss == loss and loss < 1e6, f"training diverged: loss={loss}"
-        print(json.dumps(_result(args, 1, B, elapsed, "persistent_steps_kernel", loss)))
-        return
+        if run_persistent(args.warmup):
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            assert run_persistent(args.steps)
+            torch.cuda.synchronize()
+            elapsed = time.perf_counter() - t0
+            loss = float(loss_out.item())
+            assert loss == loss and loss < 1e6, f"training diverged: loss={loss}"
+            print(json.dumps(_result(args, 1, B, elapsed, "persistent_steps_kernel", loss)))
+            return
+        print("[bench] persistent kernel declined; falling back to fused",
+              file=sys.stderr)
+        engine = "fused"
 
-    # warm up communicator + kernels, then capture one hipGraph per minibatch
+    # warm up communicator + kernels
     for i in range(3):
         eager_step((i % M) * B)
 
-    # capture G optimizer steps per graph (G | M), so one replay advances
-    # G steps — all work identical, launch overhead amortized G-fold
+    # validate capture+replay before trusting graphs in the timed loop,
+    # then capture G optimizer steps per graph (G | M) so one replay
+    # advances G steps — all work identical, launch overhead amortized
     G = max(1, args.graph_steps)
     while M % G:
         G -= 1
     graphs = None
     if use_gpu and not args.no_graph:
-        try:
-            graphs = []
-            for chunk in range(M // G):
-                g = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(g):
-                    for j in range(G):
-                        eager_step((chunk * G + j) * B)
-                graphs.append(g)
-        except RuntimeError as exc:
-            print(f"[bench] graph capture unavailable ({exc}); eager stepping",
-                  file=sys.stderr)
-            graphs = None
+        if validate_graph_capture(clf, eager_step, dist, clf.device):
+            try:
+                graphs = []
+                for chunk in range(M // G):
+                    g = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(g):
+                        for j in range(G):
+                            eager_step((chunk * G + j) * B)
+                    graphs.append(g)
+            except RuntimeError as exc:
+                print(f"[bench] graph capture unavailable ({exc}); eager stepping",
+                      file=sys.stderr)
+                graphs = None
+            # all ranks must agree on the engine (a lone eager rank would
+            # desync the RCCL collective order)
+            if dist is not None:
+                flag = torch.tensor([0.0 if graphs is None else 1.0], device=clf.device)
+                dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+                if flag.item() < 0.5:
+                    graphs = None
+        else:
+            print("[bench] graph validation failed; eager stepping", file=sys.stderr)
+
+    replay_count = [0]
 
     def run_steps(k0: int, nsteps: int):
         """Advance exactly ``nsteps`` optimizer steps from global step k0:
@@ -230,6 +457,7 @@ def main():
         while k < end:
             if graphs is not None and k % G == 0 and k + G <= end:
                 graphs[(k // G) % (M // G)].replay()
+                replay_count[0] += 1
                 k += G
             else:
                 eager_step((k % M) * B)
@@ -243,6 +471,7 @@ def main():
 
     run_steps(0, args.warmup)
     barrier_sync()
+    replay_count[0] = 0  # count only the timed region
     t0 = time.perf_counter()
     run_steps(args.warmup, args.steps)
     barrier_sync()
@@ -258,8 +487,14 @@ def main():
     assert loss == loss and loss < 1e6, f"training diverged: loss={loss}"
 
     if rank == 0:
-        engine_name = f"{engine}+{f'hipgraph{G}' if graphs is not None else 'eager'}"
-        print(json.dumps(_result(args, n_gpus, B, elapsed, engine_name, loss)))
+        # the engine label must reflect what RAN in the timed region
+        replays = replay_count[0]
+        if graphs is not None and replays > 0:
+            engine_name = f"{engine}+hipgraph{G}"
+        else:
+            engine_name = f"{engine}+eager"
+        print(json.dumps(_result(args, n_gpus, B, elapsed, engine_name, loss,
+                                 replays=replays)))
 
     if dist is not None:
         dist.destroy_process_group()

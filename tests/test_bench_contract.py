@@ -64,3 +64,63 @@ def test_bench_resnet_contract_cpu():
               "--classes", "4"])
     assert REQUIRED_FIELDS <= set(d)
     assert d["config"]["model"] == "resnet18"
+
+
+@pytest.mark.timeout(420)
+def test_bench_world4_gloo_cpu():
+    """First-contact rehearsal for the driver's 4-GPU scaling point."""
+    d = _run([
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+        "--master-port", "29591", "bench.py", "--gpus", "4", "--steps", "4",
+        "--warmup", "1", "--batch", "16", "--minibatches", "4",
+    ], timeout=400)
+    assert d["n_gpus"] == 4
+    assert d["config"]["parallelism"] == "dp4"
+    assert d["config"]["global_batch"] == 64
+
+
+@pytest.mark.timeout(600)
+def test_bench_world8_gloo_cpu():
+    """First-contact rehearsal for the driver's full-node scaling point."""
+    d = _run([
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+        "--master-port", "29592", "bench.py", "--gpus", "8", "--steps", "2",
+        "--warmup", "1", "--batch", "16", "--minibatches", "2",
+    ], timeout=560)
+    assert d["n_gpus"] == 8
+    assert d["config"]["parallelism"] == "dp8"
+
+
+@pytest.mark.timeout(300)
+def test_bench_engine_label_truthful():
+    """The engine label must say 'eager' when no graph replays happened
+    in the timed region (a hipgraphG label on an eager
+    timed region overstated what ran)."""
+    d = _run([sys.executable, "bench.py", "--steps", "4", "--warmup", "1",
+              "--batch", "32", "--minibatches", "4"])
+    assert "eager" in d["config"]["engine"]  # CPU: graphs never capture
+    assert d["config"].get("graph_replays_used", 0) == 0
+
+
+@pytest.mark.timeout(420)
+def test_bench_serve_mode_contract():
+    """--mode serve reports the /predict p50 leg of the BASELINE metric
+    through the real FastAPI app over HTTP."""
+    d = _run([sys.executable, "bench.py", "--mode", "serve", "--steps", "40",
+              "--warmup", "5"], timeout=400)
+    assert d["metric"] == "predict_p50_ms"
+    assert d["higher_is_better"] is False
+    assert d["value"] > 0
+    assert d["config"]["p99_ms"] >= d["value"]
+    assert d["config"]["rps"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_bench_persistent_engine_falls_back_cpu():
+    """Unmet persistent-engine constraints must fall back per-engine,
+    not kill the process."""
+    d = _run([sys.executable, "bench.py", "--engine", "persistent", "--steps", "4",
+              "--warmup", "1", "--batch", "48", "--minibatches", "4"])
+    assert d["value"] > 0  # survived via fallback (48 % 128 != 0, CPU anyway)
