@@ -15,6 +15,7 @@ MI355X-native additions:
   feature-selection guard — SURVEY.md §8 "Known reference quirks").
 """
 
+import inspect
 import json
 import sqlite3
 from inspect import signature
@@ -120,14 +121,14 @@ class Dataset(TrackedInstance):
         """Register the function converting reader output into the
         in-memory form the splitter/parser consume (reference:
         dataset.py:118-131)."""
-        type_guards.guard_loader(fn)
+        type_guards.guard_loader(fn, self._upstream_datatype(stage="reader"))
         self._loader = fn
         return fn
 
     def splitter(self, fn: Callable) -> Callable:
         """Register the train/test splitting function (reference:
         dataset.py:133-156)."""
-        type_guards.guard_splitter(fn)
+        type_guards.guard_splitter(fn, *self._loaded_datatype())
         self._splitter = fn
         return fn
 
@@ -135,7 +136,7 @@ class Dataset(TrackedInstance):
         """Register the function parsing a split into (features, targets, ...)
         tuples; ``feature_key`` indexes the features element (reference:
         dataset.py:158-182)."""
-        type_guards.guard_parser(fn)
+        type_guards.guard_parser(fn, self._loaded_datatype()[0])
         self._parser = fn
         self._parser_feature_key = feature_key
         return fn
@@ -158,6 +159,24 @@ class Dataset(TrackedInstance):
     # ------------------------------------------------------------------
     # derived types
     # ------------------------------------------------------------------
+
+    def _upstream_datatype(self, stage: str = "reader"):
+        """Type produced by the reader, for input-compat guards; None
+        (skip the check) when no reader is registered yet."""
+        if self._reader is None:
+            return None
+        return self.dataset_datatype["data"]
+
+    def _loaded_datatype(self):
+        """(type, source) of the data flowing out of the loader stage:
+        a custom loader's return annotation, else the reader type."""
+        if self._loader != self._default_loader:
+            ret = signature(self._loader).return_annotation
+            if ret is not inspect.Signature.empty:
+                return ret, "loader"
+            return None, "loader"
+        return self._upstream_datatype(), "reader"
+
 
     @property
     def splitter_kwargs(self) -> Dict[str, Any]:

@@ -50,6 +50,7 @@ class Model(TrackedInstance):
         *,
         init: Optional[Union[Type, Callable]] = None,
         hyperparameter_type: Optional[Type] = None,
+        hyperparameter_config: Optional[Dict[str, Type]] = None,
         dataset: Optional[Dataset] = None,
     ):
         self.name = name
@@ -58,6 +59,11 @@ class Model(TrackedInstance):
         self._dataset = dataset
         self._init_cls_or_fn: Optional[Union[Type, Callable]] = init
         self._hyperparameter_type: Optional[Type] = hyperparameter_type
+        self._hyperparameter_config: Optional[Dict[str, Type]] = hyperparameter_config
+        if hyperparameter_type is not None and hyperparameter_config is not None:
+            raise ValueError(
+                "pass either hyperparameter_type or hyperparameter_config, not both"
+            )
 
         self._init_fn: Optional[Callable] = None
         self._trainer: Optional[Callable] = None
@@ -90,6 +96,20 @@ class Model(TrackedInstance):
     def dataset(self) -> Dataset:
         return self._dataset
 
+    def _declared_model_type(self) -> Optional[type]:
+        """Model type as declared by init (class, or an init function's
+        return annotation) — None when nothing is declared, so guards
+        skip the model-type compat checks rather than comparing against
+        a guessed ``object``."""
+        if inspect.isclass(self._init_cls_or_fn):
+            return self._init_cls_or_fn
+        for fn in (self._init_fn, self._init_cls_or_fn):
+            if fn is not None:
+                ret = signature(fn).return_annotation
+                if ret is not inspect.Signature.empty and inspect.isclass(ret):
+                    return ret
+        return None
+
     @property
     def model_type(self) -> type:
         """Infer the model object's type from init/trainer annotations
@@ -114,6 +134,13 @@ class Model(TrackedInstance):
         init arg > synthesized from init keyword annotations > dict."""
         if self._hyperparameter_type is not None:
             return self._hyperparameter_type
+        if self._hyperparameter_config is not None:
+            # dict-of-types branch (reference model.py:66,180-186): the
+            # test fixtures of the reference use this path
+            return make_dataclass(
+                f"{type(self).__name__}Hyperparameters",
+                [(name, typ) for name, typ in self._hyperparameter_config.items()],
+            )
 
         init = self._init_fn or self._init_cls_or_fn
         if init is not None and not inspect.isclass(init):
@@ -199,7 +226,12 @@ class Model(TrackedInstance):
         """
 
         def decorator(f: Callable) -> Callable:
-            type_guards.guard_trainer(f, None, self._dataset.n_parser_outputs)
+            type_guards.guard_trainer(
+                f,
+                self._declared_model_type(),
+                self._dataset.n_parser_outputs,
+                self._dataset.parser_return_types,
+            )
             self._trainer = f
             self._trainer_task_kwargs = dict(
                 cache=cache, cache_version=cache_version, resources=resources, **task_kwargs
@@ -227,7 +259,9 @@ class Model(TrackedInstance):
         """Register the prediction function (reference: model.py:319-367)."""
 
         def decorator(f: Callable) -> Callable:
-            type_guards.guard_predictor(f, None)
+            type_guards.guard_predictor(
+                f, self._declared_model_type(), self._dataset.feature_type
+            )
             for cb in callbacks or []:
                 type_guards.guard_prediction_callback(cb)
             self._predictor = f
@@ -246,7 +280,12 @@ class Model(TrackedInstance):
 
     def evaluator(self, fn: Callable) -> Callable:
         """Register the evaluation function (reference: model.py:387-404)."""
-        type_guards.guard_evaluator(fn, None, self._dataset.n_parser_outputs)
+        type_guards.guard_evaluator(
+            fn,
+            self._declared_model_type(),
+            self._dataset.n_parser_outputs,
+            self._dataset.parser_return_types,
+        )
         self._evaluator = fn
         return fn
 

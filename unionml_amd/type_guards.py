@@ -69,6 +69,60 @@ def _types_compatible(expected: Any, actual: Any) -> bool:
     return exp_origin == act_origin
 
 
+def _check_input_data_type(fn_name: str, actual: Any, expected: Any) -> None:
+    """The first-argument annotation must be compatible with the type the
+    upstream stage produces (reference: type_guards.py:28-40)."""
+    if expected is None:
+        return
+    if not _types_compatible(expected, actual):
+        raise GuardError(
+            f"'{fn_name}' first argument annotated {actual} is not compatible with "
+            f"the expected input type {expected} produced by the upstream stage"
+        )
+
+
+def _is_split_container(ann: Any) -> bool:
+    origin = typing.get_origin(ann)
+    if origin in (tuple, list):
+        return True
+    # NamedTuple classes subclass tuple
+    return isinstance(ann, type) and issubclass(ann, tuple)
+
+
+def _check_split_output(fn_name: str, ann: Any, expected: Any, source: str) -> None:
+    """A splitter must return a List/Tuple/NamedTuple of the loaded data
+    type (reference: type_guards.py:43-57)."""
+    if ann is inspect.Signature.empty or ann is Any or expected is None:
+        return
+    if not _is_split_container(ann):
+        raise GuardError(
+            f"'{fn_name}' must return a List, Tuple or NamedTuple of data splits; "
+            f"annotated {ann}"
+        )
+    for sub in typing.get_args(ann):
+        if sub is Ellipsis:
+            continue
+        if not _types_compatible(expected, sub):
+            raise GuardError(
+                f"'{fn_name}' split elements annotated {sub} must match the "
+                f"'{source}' output type {expected}"
+            )
+
+
+def _check_data_args(fn_name: str, params, expected_types) -> None:
+    """Positional data arguments (after the model) must be compatible
+    with the parser's per-split return types (reference:
+    type_guards.py:118-132, 135-148)."""
+    if not expected_types:
+        return
+    for p, expected in zip(params, expected_types):
+        if not _types_compatible(expected, p.annotation):
+            raise GuardError(
+                f"'{fn_name}' data argument '{p.name}' annotated {p.annotation} is not "
+                f"compatible with the parser output type {expected}"
+            )
+
+
 def guard_reader(fn: Callable) -> None:
     """A reader may take arbitrary kwargs but MUST annotate its return
     (the return type drives downstream type derivation —
@@ -81,8 +135,9 @@ def guard_reader(fn: Callable) -> None:
         )
 
 
-def guard_loader(fn: Callable) -> None:
-    """loader(raw_data) -> loaded_data; exactly one positional argument."""
+def guard_loader(fn: Callable, expected_data_type: Any = None) -> None:
+    """loader(raw_data) -> loaded_data; exactly one positional argument
+    whose annotation must accept the reader's return type."""
     sig = _sig(fn)
     pos = _positional_params(sig)
     if len(pos) != 1:
@@ -90,9 +145,12 @@ def guard_loader(fn: Callable) -> None:
             f"loader '{_name(fn)}' must take exactly one positional argument "
             f"(the reader output), got {len(pos)}"
         )
+    _check_input_data_type(f"loader '{_name(fn)}'", pos[0].annotation, expected_data_type)
 
 
-def guard_splitter(fn: Callable) -> None:
+def guard_splitter(
+    fn: Callable, expected_data_type: Any = None, expected_type_source: str = "loader"
+) -> None:
     """splitter(data, *, test_size, shuffle, random_state) -> train/test splits."""
     sig = _sig(fn)
     pos = _positional_params(sig)
@@ -101,6 +159,11 @@ def guard_splitter(fn: Callable) -> None:
             f"splitter '{_name(fn)}' must take exactly one positional argument "
             f"(the loaded data), got {len(pos)}"
         )
+    _check_input_data_type(f"splitter '{_name(fn)}'", pos[0].annotation, expected_data_type)
+    _check_split_output(
+        f"splitter '{_name(fn)}'", sig.return_annotation, expected_data_type,
+        expected_type_source,
+    )
     kws = {p.name for p in _keyword_only_params(sig)}
     missing = set(SPLITTER_KWTYPES) - kws
     has_var_kw = any(p.kind == p.VAR_KEYWORD for p in sig.parameters.values())
@@ -111,7 +174,7 @@ def guard_splitter(fn: Callable) -> None:
         )
 
 
-def guard_parser(fn: Callable) -> None:
+def guard_parser(fn: Callable, expected_data_type: Any = None) -> None:
     """parser(data, features, targets) -> tuple of parsed outputs."""
     sig = _sig(fn)
     pos = _positional_params(sig)
@@ -120,6 +183,7 @@ def guard_parser(fn: Callable) -> None:
             f"parser '{_name(fn)}' must take three positional arguments "
             f"(data, features, targets), got {len(pos)}"
         )
+    _check_input_data_type(f"parser '{_name(fn)}'", pos[0].annotation, expected_data_type)
     if sig.return_annotation is inspect.Signature.empty:
         raise GuardError(
             f"parser '{_name(fn)}' must annotate its return type as a tuple; the "
@@ -147,7 +211,12 @@ def guard_feature_transformer(fn: Callable) -> None:
         )
 
 
-def guard_trainer(fn: Callable, model_type: Optional[type], expected_data_args: int) -> None:
+def guard_trainer(
+    fn: Callable,
+    model_type: Optional[type],
+    expected_data_args: int,
+    expected_data_types: Optional[Tuple[Any, ...]] = None,
+) -> None:
     """trainer(model, *data_args, **hyperparam/kwargs) -> model.
 
     ``expected_data_args`` is the number of elements the parser returns
@@ -164,6 +233,7 @@ def guard_trainer(fn: Callable, model_type: Optional[type], expected_data_args: 
             f"trainer '{_name(fn)}' takes {n_data} positional data argument(s) after the "
             f"model, but the dataset parser provides {expected_data_args}"
         )
+    _check_data_args(f"trainer '{_name(fn)}'", pos[1:], expected_data_types)
     if model_type is not None:
         a0 = pos[0].annotation
         if not _types_compatible(model_type, a0):
@@ -179,7 +249,12 @@ def guard_trainer(fn: Callable, model_type: Optional[type], expected_data_args: 
             )
 
 
-def guard_evaluator(fn: Callable, model_type: Optional[type], expected_data_args: int) -> None:
+def guard_evaluator(
+    fn: Callable,
+    model_type: Optional[type],
+    expected_data_args: int,
+    expected_data_types: Optional[Tuple[Any, ...]] = None,
+) -> None:
     """evaluator(model, *data_args) -> float."""
     sig = _sig(fn)
     pos = _positional_params(sig)
@@ -191,6 +266,7 @@ def guard_evaluator(fn: Callable, model_type: Optional[type], expected_data_args
             f"evaluator '{_name(fn)}' takes {n_data} positional data argument(s) after the "
             f"model, but the dataset parser provides {expected_data_args}"
         )
+    _check_data_args(f"evaluator '{_name(fn)}'", pos[1:], expected_data_types)
     if model_type is not None:
         a0 = pos[0].annotation
         if not _types_compatible(model_type, a0):
@@ -200,7 +276,9 @@ def guard_evaluator(fn: Callable, model_type: Optional[type], expected_data_args
             )
 
 
-def guard_predictor(fn: Callable, model_type: Optional[type]) -> None:
+def guard_predictor(
+    fn: Callable, model_type: Optional[type], feature_type: Any = None
+) -> None:
     """predictor(model, features) -> predictions; return annotation required."""
     sig = _sig(fn)
     pos = _positional_params(sig)
@@ -213,6 +291,11 @@ def guard_predictor(fn: Callable, model_type: Optional[type]) -> None:
         raise GuardError(
             f"predictor '{_name(fn)}' must annotate its return type: serving derives the "
             "response schema from it"
+        )
+    if feature_type is not None and not _types_compatible(feature_type, pos[1].annotation):
+        raise GuardError(
+            f"predictor '{_name(fn)}' features argument annotated {pos[1].annotation} is "
+            f"not compatible with the dataset feature type {feature_type}"
         )
     if model_type is not None:
         a0 = pos[0].annotation
