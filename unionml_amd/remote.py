@@ -82,7 +82,10 @@ def docker_build_push(image_fqn: str, dockerfile: str = "Dockerfile", context: s
         ["docker", "build", "-t", image_fqn, "-f", dockerfile, context],
         ["docker", "push", image_fqn],
     ):
-        proc = subprocess.run(cmd, capture_output=True, text=True)
+        try:
+            proc = subprocess.run(cmd, capture_output=True, text=True)
+        except FileNotFoundError as exc:
+            raise RuntimeError(f"docker is not available: {exc}") from exc
         if proc.returncode != 0:
             raise RuntimeError(f"{' '.join(cmd)} failed:\n{proc.stderr[-2000:]}")
     return image_fqn
@@ -249,6 +252,14 @@ class Backend:
                 model._trainer_task_kwargs.get("resources"), "gpu", 0
             ),
         }
+        # fast-registration code snapshot (reference remote.py:138-152
+        # zips the source and uploads it so containers run the VERSIONED
+        # code): workers import the app from this copy, so editing the
+        # live file after deploy does not change a deployed version, and
+        # patch=True re-registers code without any image build.
+        code_file = self._snapshot_code(manifest["app_module"], manifest["module_file"], app_dir)
+        if code_file:
+            manifest["code_file"] = code_file
         (app_dir / "manifest.json").write_text(json.dumps(manifest, indent=2))
 
         import cloudpickle
@@ -259,13 +270,45 @@ class Backend:
         if self.registry and not patch:
             image_fqn = get_image_fqn(self.registry, self.image_name, model.name, app_version)
             os.environ["UNIONML_INTERNAL_IMAGE"] = image_fqn
-            try:
-                docker_build_push(image_fqn, self.dockerfile)
-            except (RuntimeError, FileNotFoundError) as exc:
-                logger.warning("docker packaging skipped: %s", exc)
+            # an explicitly configured registry means the user asked for
+            # an image: a failed build/push is a deploy failure, not a
+            # warning to swallow
+            docker_build_push(image_fqn, self.dockerfile)
 
-        logger.info("deployed app version %s to %s", app_version, app_dir)
+        logger.info(
+            "deployed app version %s to %s%s", app_version, app_dir,
+            " (patch: code-only)" if patch else "",
+        )
         return app_version
+
+    def _snapshot_code(self, app_module: str, module_file: Optional[str], app_dir: Path):
+        """Copy the app's source into the version dir so workers run the
+        registered code. Framework-internal apps (``unionml_amd.*``) are
+        provided by the installed framework (the reference's analog: the
+        container image provides the library; fast-registration ships
+        only user code)."""
+        import shutil
+
+        if not module_file or app_module.split(".")[0] == "unionml_amd":
+            return None
+        mf = Path(module_file)
+        if not mf.exists():
+            return None
+        code_dir = app_dir / "code"
+        if code_dir.exists():
+            shutil.rmtree(code_dir)
+        code_dir.mkdir(parents=True)
+        parts = app_module.split(".")
+        if len(parts) == 1:
+            shutil.copy2(mf, code_dir / mf.name)
+            return str(Path("code") / mf.name)
+        # module inside a package: copy the whole root package
+        root_pkg = mf.parents[len(parts) - 2]
+        shutil.copytree(
+            root_pkg, code_dir / root_pkg.name,
+            ignore=shutil.ignore_patterns("__pycache__", ".git", "*.so"),
+        )
+        return str(Path("code") / root_pkg.name / Path(*parts[1:]).with_suffix(".py"))
 
     def latest_app_version(self) -> Optional[str]:
         apps = sorted(
@@ -275,13 +318,18 @@ class Backend:
         return apps[-1].name if apps else None
 
     def _manifest(self, app_version: Optional[str]) -> Dict:
+        manifest, _ = self._manifest_with_dir(app_version)
+        return manifest
+
+    def _manifest_with_dir(self, app_version: Optional[str]):
         app_version = app_version or self.latest_app_version()
         if app_version is None:
             raise ModelArtifactNotFound(f"no app deployed in project '{self.project}'")
-        path = self.root / "apps" / app_version / "manifest.json"
+        app_dir = self.root / "apps" / app_version
+        path = app_dir / "manifest.json"
         if not path.exists():
             raise ModelArtifactNotFound(f"app version '{app_version}' is not deployed")
-        return json.loads(path.read_text())
+        return json.loads(path.read_text()), app_dir
 
     # ------------------------------------------------------------------
     # execute
@@ -301,11 +349,18 @@ class Backend:
         import cloudpickle
 
         try:
-            manifest = self._manifest(app_version)
+            manifest, app_dir = self._manifest_with_dir(app_version)
         except ModelArtifactNotFound:
             # auto-deploy for local-first ergonomics
             self.deploy(model, allow_uncommitted=True)
-            manifest = self._manifest(None)
+            manifest, app_dir = self._manifest_with_dir(None)
+
+        # prefer the fast-registration code snapshot: the worker runs the
+        # REGISTERED code, not whatever the live tree looks like now
+        module_file = manifest.get("module_file")
+        code_file = manifest.get("code_file")
+        if code_file and (app_dir / code_file).exists():
+            module_file = str(app_dir / code_file)
 
         exec_id = f"{workflow}-{datetime.datetime.now():%Y%m%d%H%M%S}-{uuid.uuid4().hex[:6]}"
         exec_dir = self.root / "executions" / exec_id
@@ -315,7 +370,7 @@ class Backend:
             "workflow": workflow,
             "app_module": manifest["app_module"],
             "object_name": manifest["object_name"],
-            "module_file": manifest.get("module_file"),
+            "module_file": module_file,
             "schedule_name": schedule_name,
             "model_name": manifest["model_name"],
         }
@@ -335,7 +390,6 @@ class Backend:
         import unionml_amd
 
         pythonpath = [str(Path(unionml_amd.__file__).parent.parent)]
-        module_file = manifest.get("module_file")
         if module_file:
             pythonpath.insert(0, str(Path(module_file).parent))
         if env.get("PYTHONPATH"):
