@@ -49,10 +49,8 @@ def evaluator(
 fastapi_app = FastAPI()
 model.serve(fastapi_app)
 
-# the Lambda entrypoint: handler = Mangum-wrapped ASGI app
-try:
-    from mangum import Mangum
+# the Lambda entrypoint: Mangum when installed, else the built-in
+# dependency-free API-Gateway adapter
+from unionml_amd.services.awslambda import lambda_handler_for
 
-    lambda_handler = Mangum(fastapi_app)
-except ImportError:  # mangum only needed inside the Lambda package
-    lambda_handler = None
+lambda_handler = lambda_handler_for(fastapi_app)
