@@ -37,7 +37,6 @@ def serving_app(
         except ImportError:  # prometheus_client not installed
             pass
 
-    @app.on_event("startup")
     async def load_model():
         # local: $UNIONML_MODEL_PATH; remote: latest successful training
         # run from the backend registry (reference: fastapi.py:22-34)
@@ -56,10 +55,26 @@ def serving_app(
             )
             state["batcher"].start()
 
-    @app.on_event("shutdown")
     async def stop_batcher():
         if state["batcher"] is not None:
             state["batcher"].stop()
+
+    # wrap (don't replace) any lifespan the user's app already has —
+    # the modern replacement for the deprecated @app.on_event hooks
+    from contextlib import asynccontextmanager
+
+    existing_lifespan = app.router.lifespan_context
+
+    @asynccontextmanager
+    async def lifespan(app_):
+        async with existing_lifespan(app_):
+            await load_model()
+            try:
+                yield
+            finally:
+                await stop_batcher()
+
+    app.router.lifespan_context = lifespan
 
     @app.get("/")
     async def root():
