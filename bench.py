@@ -73,6 +73,8 @@ def parse_args():
     )
     p.add_argument("--serve-clients", type=int, default=1,
                    help="concurrent HTTP clients in serve mode")
+    p.add_argument("--serve-workers", type=int, default=1,
+                   help="SO_REUSEPORT server workers in serve mode")
     return p.parse_args()
 
 
@@ -108,6 +110,46 @@ def _result(args, n_gpus, B, elapsed, engine, loss, replays=None):
 # serve mode: boot the real FastAPI app + uvicorn, measure /predict p50
 # ---------------------------------------------------------------------------
 
+def _client_loop(port, body, n_requests, out_lats, barrier=None):
+    """One load client: a persistent keep-alive connection issuing POSTs
+    (new-connection-per-request clients measure TCP setup, not the
+    server)."""
+    import http.client
+
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+    headers = {"Content-Type": "application/json", "Connection": "keep-alive"}
+    if barrier is not None:
+        barrier.wait()
+    for _ in range(n_requests):
+        t0 = time.perf_counter()
+        conn.request("POST", "/predict", body=body, headers=headers)
+        resp = conn.getresponse()
+        resp.read()
+        if resp.status != 200:  # count, don't kill the client thread
+            out_lats.append(float("nan"))
+            continue
+        out_lats.append((time.perf_counter() - t0) * 1000.0)
+    conn.close()
+
+
+def _client_proc(port, body, n_threads, n_requests_per_thread, q, start_evt):
+    import threading
+
+    per = [[] for _ in range(n_threads)]
+    barrier = threading.Barrier(n_threads)
+
+    def run(i):
+        start_evt.wait()
+        _client_loop(port, body, n_requests_per_thread, per[i], barrier)
+
+    threads = [threading.Thread(target=run, args=(i,)) for i in range(n_threads)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    q.put([x for sub in per for x in sub])
+
+
 def run_serve_mode(args):
     import socket
     import subprocess
@@ -132,50 +174,80 @@ def run_serve_mode(args):
 
     env = dict(os.environ)
     env["UNIONML_MODEL_PATH"] = artifact_path
-    server = subprocess.Popen(
-        [sys.executable, "-m", "uvicorn", "--host", "127.0.0.1", "--port", str(port),
-         "--log-level", "warning", "unionml_amd.models.mlp_serve:app"],
-        env=env,
-    )
+    if args.serve_workers > 1:
+        # the framework's SO_REUSEPORT multi-worker supervisor (each
+        # worker owns its bucketed inference hipGraphs)
+        server = subprocess.Popen(
+            [sys.executable, "-m", "unionml_amd.cli", "serve",
+             "unionml_amd.models.mlp_serve:app", "--host", "127.0.0.1",
+             "--port", str(port), "--workers", str(args.serve_workers)],
+            env=env,
+        )
+    else:
+        server = subprocess.Popen(
+            [sys.executable, "-m", "uvicorn", "--host", "127.0.0.1", "--port",
+             str(port), "--log-level", "warning", "unionml_amd.models.mlp_serve:app"],
+            env=env,
+        )
     try:
-        # wait for readiness
+        # wait for readiness: behind SO_REUSEPORT each worker loads its
+        # artifact and captures its own hipGraphs independently, so
+        # demand a long streak of consecutive healthy responses (enough
+        # to have hit every worker with high probability)
         rng_rows = [{f"p{i}": float((i * 7) % 16) for i in range(64)}]
         body = json.dumps({"features": rng_rows}).encode()
-        deadline = time.time() + 120
-        while True:
+        deadline = time.time() + 180
+        streak, need = 0, 16 * max(1, args.serve_workers)
+        while streak < need:
             try:
                 urllib.request.urlopen(f"http://127.0.0.1:{port}/health", timeout=2)
-                break
+                streak += 1
             except Exception:
+                streak = 0
                 if time.time() > deadline:
                     raise RuntimeError("serve-mode server never became healthy")
                 time.sleep(0.5)
 
-        def one_request():
-            req = urllib.request.Request(
-                f"http://127.0.0.1:{port}/predict", data=body,
-                headers={"Content-Type": "application/json"},
-            )
-            t0 = time.perf_counter()
-            with urllib.request.urlopen(req, timeout=10) as resp:
-                resp.read()
-            return (time.perf_counter() - t0) * 1000.0
-
-        for _ in range(warmup):
-            one_request()
+        warm = []
+        _client_loop(port, body, warmup, warm)
         lats = []
         if args.serve_clients <= 1:
             t_start = time.perf_counter()
-            for _ in range(steps):
-                lats.append(one_request())
+            _client_loop(port, body, steps, lats)
             elapsed = time.perf_counter() - t_start
         else:
-            import concurrent.futures as cf
+            # fan the load clients across PROCESSES (a single GIL-bound
+            # client process inflates tail latencies at high rps)
+            import multiprocessing as mp
 
+            n_clients = args.serve_clients
+            n_procs = min(n_clients, 8)
+            threads_per = n_clients // n_procs
+            per_thread = steps // (n_procs * threads_per)
+            ctx = mp.get_context("fork")
+            q = ctx.Queue()
+            start_evt = ctx.Event()
+            procs = [
+                ctx.Process(
+                    target=_client_proc,
+                    args=(port, body, threads_per, per_thread, q, start_evt),
+                )
+                for _ in range(n_procs)
+            ]
+            for p_ in procs:
+                p_.start()
+            time.sleep(0.5)  # let every process build its connections
             t_start = time.perf_counter()
-            with cf.ThreadPoolExecutor(max_workers=args.serve_clients) as pool:
-                lats = list(pool.map(lambda _: one_request(), range(steps)))
+            start_evt.set()
+            lats = []
+            for _ in procs:
+                lats.extend(q.get())
+            for p_ in procs:
+                p_.join()
             elapsed = time.perf_counter() - t_start
+        errors = sum(1 for x in lats if x != x)
+        lats = [x for x in lats if x == x]
+        steps = len(lats)
         lats.sort()
         p = lambda q: lats[min(len(lats) - 1, int(q * len(lats)))]  # noqa: E731
         print(json.dumps({
@@ -197,9 +269,11 @@ def run_serve_mode(args):
                 "seq_len": None,
                 "parallelism": "serve1",
                 "clients": args.serve_clients,
+                "workers": args.serve_workers,
                 "p90_ms": p(0.90),
                 "p99_ms": p(0.99),
                 "rps": steps / elapsed,
+                "errors": errors,
             },
         }))
     finally:

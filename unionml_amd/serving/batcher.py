@@ -143,6 +143,20 @@ class DynamicBatcher:
                 batch.append(self._queue.popleft())
             return batch
 
+    def _numeric_fast_path_ok(self) -> bool:
+        """True when requests can bypass the pandas feature pipeline:
+        the forward is a graphed runner (consumes a float32 matrix) and
+        the dataset uses the default feature loader/transformer with
+        declared feature columns — the per-request DataFrame build is
+        pure overhead on the /predict hot path then."""
+        ds = self.model._dataset
+        return (
+            self._graphed is not None
+            and ds._feature_loader == ds._default_feature_loader
+            and ds._feature_transformer == ds._default_feature_transformer
+            and bool(ds._features)
+        )
+
     def _predict_batch(self, features_list: List[Any]):
         """One forward over the concatenated features of the batch."""
         import numpy as np
@@ -150,6 +164,27 @@ class DynamicBatcher:
 
         model = self.model
         ds = model._dataset
+
+        if self._numeric_fast_path_ok():
+            names = ds._features
+            rows: List[Any] = []
+            counts = []
+            try:
+                for f in features_list:
+                    if not (isinstance(f, list) and f and isinstance(f[0], dict)):
+                        raise TypeError  # non-records request: general path
+                    counts.append(len(f))
+                    rows.extend([rec[c] for c in names] for rec in f)
+                merged = np.asarray(rows, dtype=np.float32)
+            except (TypeError, KeyError, ValueError):
+                pass
+            else:
+                predictions = self._forward(merged)
+                out, offset = [], 0
+                for n in counts:
+                    out.append(predictions[offset : offset + n])
+                    offset += n
+                return out
         loaded = [ds.get_features(f) for f in features_list]
         first = loaded[0]
         if isinstance(first, pd.DataFrame):
