@@ -122,10 +122,19 @@ def _client_loop(port, body, n_requests, out_lats, barrier=None):
         barrier.wait()
     for _ in range(n_requests):
         t0 = time.perf_counter()
-        conn.request("POST", "/predict", body=body, headers=headers)
-        resp = conn.getresponse()
-        resp.read()
-        if resp.status != 200:  # count, don't kill the client thread
+        try:
+            conn.request("POST", "/predict", body=body, headers=headers)
+            resp = conn.getresponse()
+            resp.read()
+            status = resp.status
+        except (OSError, http.client.HTTPException):
+            # connection reset (e.g. a worker restarting): reconnect,
+            # count the failure, keep the client alive
+            conn.close()
+            conn = http.client.HTTPConnection("127.0.0.1", port, timeout=10)
+            out_lats.append(float("nan"))
+            continue
+        if status != 200:  # count, don't kill the client thread
             out_lats.append(float("nan"))
             continue
         out_lats.append((time.perf_counter() - t0) * 1000.0)
