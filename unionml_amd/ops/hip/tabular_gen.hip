@@ -83,9 +83,37 @@ struct GG {
   static constexpr int W2S = 40;        // W2s       [HID][40]  (K-pad to 32)
   static constexpr int W2T = HID + 8;   // W2T       [CPAD][W2T]
 
-  static constexpr int O_XB = 0;  // xbuf: Xs tiles (fwd) then XT tiles (bwd)
-  static constexpr int O_W1 = A16(O_XB + IMAX(RT * XSS, TK * XTS) * 2);
-  static constexpr int O_HS = A16(O_W1 + HID * W1S * 2);
+  // streamed-tile buffer sizes (u16 elems, ONE buffer each)
+  static constexpr int XBUF1 = IMAX(RT * XSS, TK * XTS);
+  static constexpr int W1B1 = HID * W1S;
+
+  static constexpr int layout_total(int nxb) {
+    int o = A16(nxb * XBUF1 * 2);
+    o = A16(o + nxb * W1B1 * 2);
+    o = A16(o + RT * HSS * 2);
+    o = A16(o + HID * HTS * 2);
+    o = A16(o + RT * DLS * 2);
+    o = A16(o + CPAD * DTS * 2);
+    o = A16(o + HID * DHS * 2);
+    o = A16(o + HID * W2S * 2);
+    o = A16(o + CPAD * W2T * 2);
+    o = A16(o + HID * 4);
+    o = A16(o + CPAD * 4);
+    o = A16(o + 16);
+    o = A16(o + HID * 4);
+    return A16(o + CPAD * 4);
+  }
+
+  // double-buffer the k-streamed tiles (Xs/W1T in fwd1, XT in dW1)
+  // whenever the carve still fits in the CU's 160 KB: the next tile's
+  // cooperative load then overlaps the current tile's MFMAs, hiding
+  // the L2/HBM round trip that otherwise serializes each k step
+  static constexpr bool DB = layout_total(2) <= 160 * 1024;
+  static constexpr int NXB = DB ? 2 : 1;
+
+  static constexpr int O_XB = 0;
+  static constexpr int O_W1 = A16(O_XB + NXB * XBUF1 * 2);
+  static constexpr int O_HS = A16(O_W1 + NXB * W1B1 * 2);
   static constexpr int O_HT = A16(O_HS + RT * HSS * 2);
   static constexpr int O_DL = A16(O_HT + HID * HTS * 2);
   static constexpr int O_DT = A16(O_DL + RT * DLS * 2);
@@ -98,6 +126,7 @@ struct GG {
   static constexpr int O_B1 = A16(O_LS + 16);                // b1 stage [HID]
   static constexpr int O_B2 = A16(O_B1 + HID * 4);           // b2 stage [CPAD]
   static constexpr int TOTAL = A16(O_B2 + CPAD * 4);
+  static_assert(TOTAL == layout_total(NXB), "layout helper out of sync");
 
   static constexpr int NT1 = (HID / 16) * (RT / 16);  // fwd1 / dH D tiles
   static constexpr int SLOTS = (NT1 + WAVES - 1) / WAVES;
@@ -106,9 +135,7 @@ struct GG {
   static_assert(RT % 32 == 0, "RT must be a multiple of 32");
   static_assert(TOTAL <= 160 * 1024, "LDS carve exceeds 160 KB");
 
-  u16 (*Xs)[XSS];
-  u16 (*XT)[XTS];
-  u16 (*W1T)[W1S];
+  char* smem_;
   u16 (*Hs)[HSS];
   u16 (*HT)[HTS];
   u16 (*DLs)[DLS];
@@ -122,10 +149,18 @@ struct GG {
   float* b1s;
   float* b2s;
 
+  __device__ __forceinline__ u16 (*Xs(int b))[XSS] {
+    return (u16(*)[XSS])(smem_ + O_XB + b * XBUF1 * 2);
+  }
+  __device__ __forceinline__ u16 (*XT(int b))[XTS] {
+    return (u16(*)[XTS])(smem_ + O_XB + b * XBUF1 * 2);
+  }
+  __device__ __forceinline__ u16 (*W1T(int b))[W1S] {
+    return (u16(*)[W1S])(smem_ + O_W1 + b * W1B1 * 2);
+  }
+
   __device__ __forceinline__ void carve(char* smem) {
-    Xs = (u16(*)[XSS])(smem + O_XB);
-    XT = (u16(*)[XTS])(smem + O_XB);
-    W1T = (u16(*)[W1S])(smem + O_W1);
+    smem_ = smem;
     Hs = (u16(*)[HSS])(smem + O_HS);
     HT = (u16(*)[HTS])(smem + O_HT);
     DLs = (u16(*)[DLS])(smem + O_DL);
@@ -172,9 +207,11 @@ __device__ __forceinline__ void wimg_write_gen(u16* __restrict__ wimg, int inp,
 
 // Xs tile: rows of this WG's chunk, k-columns [k0, k0+kv), zero tail
 template <typename G>
-__device__ __forceinline__ void load_xs_tile(const G& L, const u16* __restrict__ Xbf,
+__device__ __forceinline__ void load_xs_tile(G& L, int buf,
+                                             const u16* __restrict__ Xbf,
                                              int inp, long long row0,
                                              long long Nvalid, int k0, int kv) {
+  u16 (*Xs)[G::XSS] = L.Xs(buf);
   for (int i = threadIdx.x; i < G::RT * (G::TK / 8); i += G::BLOCK) {
     const int r = i / (G::TK / 8);
     const int jg = (i % (G::TK / 8)) * 8;
@@ -182,15 +219,17 @@ __device__ __forceinline__ void load_xs_tile(const G& L, const u16* __restrict__
     if (row0 + r < Nvalid && jg < kv) {
       v = *(const bf16x8*)&Xbf[(row0 + r) * inp + k0 + jg];
     }
-    *(bf16x8*)&L.Xs[r][jg] = v;
+    *(bf16x8*)&Xs[r][jg] = v;
   }
 }
 
 // XT tile: transposed image of the same chunk/k-tile (dW1's B operand)
 template <typename G>
-__device__ __forceinline__ void load_xt_tile(const G& L, const u16* __restrict__ Xbf,
+__device__ __forceinline__ void load_xt_tile(G& L, int buf,
+                                             const u16* __restrict__ Xbf,
                                              int inp, long long row0,
                                              long long Nvalid, int k0, int kv) {
+  u16 (*XT)[G::XTS] = L.XT(buf);
   for (int i = threadIdx.x; i < G::RT * (G::TK / 8); i += G::BLOCK) {
     const int r = i / (G::TK / 8);
     const int jg = (i % (G::TK / 8)) * 8;
@@ -199,20 +238,22 @@ __device__ __forceinline__ void load_xt_tile(const G& L, const u16* __restrict__
       v = *(const bf16x8*)&Xbf[(row0 + r) * inp + k0 + jg];
     }
     #pragma unroll
-    for (int j = 0; j < 8; ++j) L.XT[jg + j][r] = (u16)v[j];
+    for (int j = 0; j < 8; ++j) XT[jg + j][r] = (u16)v[j];
   }
 }
 
 // W1T k-tile from the packed global image (row-major [HID][inp])
 template <typename G>
-__device__ __forceinline__ void load_w1t_tile(const G& L, const u16* __restrict__ wimg,
+__device__ __forceinline__ void load_w1t_tile(G& L, int buf,
+                                              const u16* __restrict__ wimg,
                                               int inp, int k0, int kv) {
+  u16 (*W1T)[G::W1S] = L.W1T(buf);
   for (int i = threadIdx.x; i < G::HID * (G::TK / 8); i += G::BLOCK) {
     const int h = i / (G::TK / 8);
     const int jg = (i % (G::TK / 8)) * 8;
     bf16x8 v = (bf16x8){0, 0, 0, 0, 0, 0, 0, 0};
     if (jg < kv) v = *(const bf16x8*)&wimg[h * inp + k0 + jg];
-    *(bf16x8*)&L.W1T[h][jg] = v;
+    *(bf16x8*)&W1T[h][jg] = v;
   }
 }
 
@@ -249,27 +290,20 @@ __global__ void __launch_bounds__(512)
 mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero-padded)
                     const int* __restrict__ y, int B,
                     int inp, int cls,
-                    u16* __restrict__ wimg,        // packed weight images
-                    float* __restrict__ master, u16* __restrict__ bfmirror,
-                    float* __restrict__ m, float* __restrict__ v,
-                    int* __restrict__ t_dev,
+                    const u16* __restrict__ wimg,  // packed weight images
+                    const float* __restrict__ master,  // biases prefetch
                     float* __restrict__ slabs, int slab_stride,
-                    unsigned* __restrict__ counter,
-                    float* __restrict__ loss_out,
-                    float invBtot, float lr, float beta1, float beta2, float eps,
-                    float* __restrict__ grads_out) {
+                    float invBtot) {
   using G = GG<HID, RT>;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   G L;
   L.carve(smem);
-  unsigned* lossu = (unsigned*)L.loss;
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int l = tid & 63;
   const int lg = l >> 4, lr_ = l & 15;
   const long long row0 = (long long)blockIdx.x * RT;
-  const int n_wg = gridDim.x;
 
   const int off_b1 = inp * HID;
   const int off_w2 = off_b1 + HID;
@@ -278,31 +312,34 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
 
   for (int i = tid; i < HID; i += G::BLOCK) L.db1[i] = 0.f;
   if (tid < G::CPAD) L.db2[tid] = 0.f;
-  if (tid == 0) {
-    L.loss[0] = 0.f;
-    // epoch/base reads BEFORE any WG of this launch can advance them
-    // (writer runs only after every WG published — see G16 notes in
-    // tabular_kernels.hip)
-    lossu[2] = *counter;
-    lossu[3] = (unsigned)(*t_dev);
-  }
+  if (tid == 0) L.loss[0] = 0.f;
   zero_dl_pad(L);
   load_w2_images(L, wimg, inp);
   for (int i = tid; i < HID; i += G::BLOCK) L.b1s[i] = master[off_b1 + i];
   if (tid < G::CPAD) L.b2s[tid] = master[off_b2 + tid];
   __syncthreads();
 
-  // ---- fwd1: H^T = W1T @ B(Xs), K streamed in TK tiles over inp ------------
+  // ---- fwd1: H^T = W1T @ B(Xs), K streamed in TK tiles over inp.
+  // Double-buffered where LDS allows (G::DB): the NEXT tile's
+  // cooperative load issues before this tile's MFMAs, so the L2/HBM
+  // round trip overlaps compute instead of serializing each k step.
   f32x4 acc1[G::SLOTS];
   #pragma unroll
   for (int s = 0; s < G::SLOTS; ++s) acc1[s] = (f32x4){0.f, 0.f, 0.f, 0.f};
   const int nkt = (inp + G::TK - 1) / G::TK;
+  load_xs_tile(L, 0, Xbf, inp, row0, (long long)B, 0, min(G::TK, inp));
+  load_w1t_tile(L, 0, wimg, inp, 0, min(G::TK, inp));
+  __syncthreads();
   for (int kt = 0; kt < nkt; ++kt) {
-    const int k0 = kt * G::TK;
-    const int kv = min(G::TK, inp - k0);
-    load_xs_tile(L, Xbf, inp, row0, (long long)B, k0, kv);
-    load_w1t_tile(L, wimg, inp, k0, kv);
-    __syncthreads();
+    const int cur = G::DB ? (kt & 1) : 0;
+    if (G::DB && kt + 1 < nkt) {
+      const int k0n = (kt + 1) * G::TK;
+      const int kvn = min(G::TK, inp - k0n);
+      load_xs_tile(L, cur ^ 1, Xbf, inp, row0, (long long)B, k0n, kvn);
+      load_w1t_tile(L, cur ^ 1, wimg, inp, k0n, kvn);
+    }
+    u16 (*W1Tc)[G::W1S] = L.W1T(cur);
+    u16 (*Xsc)[G::XSS] = L.Xs(cur);
     #pragma unroll
     for (int s = 0; s < G::SLOTS; ++s) {
       const int t = wave + G::WAVES * s;
@@ -312,14 +349,21 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
         f32x4 acc = acc1[s];
         #pragma unroll
         for (int ks = 0; ks < G::TK / 32; ++ks) {
-          const bf16x8 a = *(const bf16x8*)&L.W1T[mt * 16 + lr_][ks * 32 + lg * 8];
-          const bf16x8 b = *(const bf16x8*)&L.Xs[rt * 16 + lr_][ks * 32 + lg * 8];
+          const bf16x8 a = *(const bf16x8*)&W1Tc[mt * 16 + lr_][ks * 32 + lg * 8];
+          const bf16x8 b = *(const bf16x8*)&Xsc[rt * 16 + lr_][ks * 32 + lg * 8];
           acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
         }
         acc1[s] = acc;
       }
     }
-    __syncthreads();  // tile buffers free before the next k-tile load
+    __syncthreads();  // prefetched buffer complete / current reads done
+    if (!G::DB && kt + 1 < nkt) {
+      const int k0n = (kt + 1) * G::TK;
+      const int kvn = min(G::TK, inp - k0n);
+      load_xs_tile(L, 0, Xbf, inp, row0, (long long)B, k0n, kvn);
+      load_w1t_tile(L, 0, wimg, inp, k0n, kvn);
+      __syncthreads();
+    }
   }
   // epilogue: bias + relu, store Hs and HT
   #pragma unroll
@@ -451,12 +495,20 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
   }
 
   // ---- dW1^T = DHT @ B(XT), K'-streamed over inp -> slab -------------------
+  // (xbuf is free: fwd1's last Xs read completed before its loop-end
+  // barrier; dW2 above touches only HT/DLT)
+  load_xt_tile(L, 0, Xbf, inp, row0, (long long)B, 0, min(G::TK, inp));
+  __syncthreads();
   for (int kt = 0; kt < nkt; ++kt) {
+    const int cur = G::DB ? (kt & 1) : 0;
+    if (G::DB && kt + 1 < nkt) {
+      const int k0n = (kt + 1) * G::TK;
+      load_xt_tile(L, cur ^ 1, Xbf, inp, row0, (long long)B, k0n,
+                   min(G::TK, inp - k0n));
+    }
     const int k0 = kt * G::TK;
     const int kv = min(G::TK, inp - k0);
-    __syncthreads();  // xbuf free (fwd done); also previous XT tile consumed
-    load_xt_tile(L, Xbf, inp, row0, (long long)B, k0, kv);
-    __syncthreads();
+    u16 (*XTc)[G::XTS] = L.XT(cur);
     const int ntw = (HID / 16) * (kv / 16);
     for (int t = wave; t < ntw; t += G::WAVES) {
       const int mt = t % (HID / 16);
@@ -465,7 +517,7 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
       #pragma unroll
       for (int ks = 0; ks < RT / 32; ++ks) {
         const bf16x8 a = *(const bf16x8*)&L.DHT[mt * 16 + lr_][ks * 32 + lg * 8];
-        const bf16x8 b = *(const bf16x8*)&L.XT[it * 16 + lr_][ks * 32 + lg * 8];
+        const bf16x8 b = *(const bf16x8*)&XTc[it * 16 + lr_][ks * 32 + lg * 8];
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
       }
       #pragma unroll
@@ -475,76 +527,67 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
         slab[in * HID + h] = acc[r];
       }
     }
+    __syncthreads();
+    if (!G::DB && kt + 1 < nkt) {
+      const int k0n = (kt + 1) * G::TK;
+      load_xt_tile(L, 0, Xbf, inp, row0, (long long)B, k0n,
+                   min(G::TK, inp - k0n));
+      __syncthreads();
+    }
   }
-  __syncthreads();
 
   // ---- bias grads + loss into the slab -------------------------------------
   for (int i = tid; i < HID; i += G::BLOCK) slab[off_b1 + i] = L.db1[i];
   if (tid < G::CPAD) slab[off_b2 + tid] = L.db2[tid];
   if (tid == G::CPAD) slab[nparam] = L.loss[0];
 
-  // ---- publish + all-WG barrier (G16: release + per-slab epoch tag) --------
-  const unsigned epoch = lossu[2] + 1u;
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-  if (tid == 0) {
-    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __hip_atomic_store((unsigned*)(slab + nparam + 1), epoch, __ATOMIC_RELAXED,
-                       __HIP_MEMORY_SCOPE_AGENT);
-  }
-  __syncthreads();
+}
 
-  if (wave == 0) {
-    unsigned spins = 0;
-    bool ok = true;
-    for (int w = l; w < n_wg; w += 64) {
-      for (;;) {
-        const unsigned tag = __hip_atomic_load(
-            (const unsigned*)(slabs + (long long)w * slab_stride + nparam + 1),
-            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-        if (tag == epoch) break;
-        __builtin_amdgcn_s_sleep(2);
-        if (++spins > 100000000u) { ok = false; break; }
-      }
-      if (!ok) break;
-    }
-    const unsigned long long bad = __ballot(!ok);
-    if (l == 0) {
-      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-      lossu[1] = bad ? 1u : 0u;
-    }
-  }
-  __syncthreads();
-  if (lossu[1] != 0u) {  // timed out: poison the loss, keep going
-    if (tid == 0 && blockIdx.x == 0) *loss_out = __builtin_nanf("");
-    return;
-  }
+// ---------------------------------------------------------------------------
+// wide-grid reduce + Adam: sums the per-WG slabs and applies Adam (or,
+// in grads_out mode, hands the summed grads to the RCCL all-reduce).
+//
+// Design note vs the specialized kernel's single-launch handshake
+// (tabular_kernels.hip G16): at generalized geometries nparam is large
+// (MNIST shape: ~102k params -> ~8 MB of slab+state traffic per step)
+// and the step grid is tiny (ceil(B/RT) WGs), so an in-kernel reduce is
+// bandwidth-starved — memory parallelism scales with resident CUs. A
+// separate kernel launches with hundreds of WGs, reaches full-chip
+// bandwidth, and the kernel boundary IS the inter-WG barrier, so the
+// epoch-tag handshake disappears. Stream-ordered + hipGraph-capturable;
+// t advances via bump_t_kernel so every WG sees one consistent t.
+// ---------------------------------------------------------------------------
 
-  // ---- every WG reduces its own param stripe + applies Adam ----------------
-  const float t_new = (float)(lossu[3] + 1u);
+__global__ void __launch_bounds__(256)
+reduce_adam_gen_kernel(const float* __restrict__ slabs, int n_wg,
+                       int slab_stride, int nparam, int inp, int hid,
+                       float* __restrict__ master, u16* __restrict__ bfmirror,
+                       float* __restrict__ m, float* __restrict__ v,
+                       int* __restrict__ t_dev,
+                       float* __restrict__ loss_out,
+                       float lr, float beta1, float beta2, float eps,
+                       u16* __restrict__ wimg,
+                       float* __restrict__ grads_out,
+                       unsigned* __restrict__ done_counter) {
+  const float t_new = (float)(*t_dev + 1);
   const float corr1 = fast_rcp(1.f - __powf(beta1, t_new));
   const float corr2 = fast_rcp(1.f - __powf(beta2, t_new));
-  const int span = (nparam + 1 + n_wg - 1) / n_wg;
-  const int lo = blockIdx.x * span;
-  const int hi = min(lo + span, nparam + 1);
-  for (int i = lo + tid; i < hi; i += G::BLOCK) {
+  const int off_b1 = inp * hid;
+  const int off_w2 = off_b1 + hid;
+  const int off_b2 = off_w2 + hid * 16;
+  for (int i = blockIdx.x * 256 + threadIdx.x; i <= nparam;
+       i += gridDim.x * 256) {
     float g0 = 0.f, g1 = 0.f, g2 = 0.f, g3 = 0.f;
-    float g4 = 0.f, g5 = 0.f, g6 = 0.f, g7 = 0.f;
     int w = 0;
-    for (; w + 8 <= n_wg; w += 8) {
+    for (; w + 4 <= n_wg; w += 4) {
       g0 += slabs[(long long)w * slab_stride + i];
       g1 += slabs[(long long)(w + 1) * slab_stride + i];
       g2 += slabs[(long long)(w + 2) * slab_stride + i];
       g3 += slabs[(long long)(w + 3) * slab_stride + i];
-      g4 += slabs[(long long)(w + 4) * slab_stride + i];
-      g5 += slabs[(long long)(w + 5) * slab_stride + i];
-      g6 += slabs[(long long)(w + 6) * slab_stride + i];
-      g7 += slabs[(long long)(w + 7) * slab_stride + i];
     }
     for (; w < n_wg; ++w) g0 += slabs[(long long)w * slab_stride + i];
-    const float g = ((g0 + g1) + (g2 + g3)) + ((g4 + g5) + (g6 + g7));
-    if (grads_out) {        // reduce-only: hand summed grads to RCCL
+    const float g = (g0 + g1) + (g2 + g3);
+    if (grads_out) {          // reduce-only: RCCL takes it from here
       grads_out[i] = g;
       continue;
     }
@@ -560,11 +603,27 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
     master[i] = p;
     const u16 wb = f2bf(p);
     bfmirror[i] = wb;
-    wimg_write_gen<HID>(wimg, inp, off_b1, off_w2, off_b2, i, wb);
+    if (wimg) {
+      if (i < off_b1) {
+        wimg[(i % hid) * inp + (i / hid)] = wb;
+      } else if (i >= off_w2 && i < off_b2) {
+        const int j = i - off_w2;
+        const int h = j >> 4, c = j & 15;
+        wimg[hid * inp + h * 32 + c] = wb;
+        wimg[hid * inp + hid * 32 + c * hid + h] = wb;
+      }
+    }
   }
-  if (tid == 0 && blockIdx.x == 0) {
-    *counter = epoch;
-    if (!grads_out) *t_dev = (int)t_new;
+  // advance the Adam step counter in-kernel: every WG read t_dev at
+  // entry (the last-done WG writes only after all WGs passed their
+  // read), saving the separate 1-thread bump kernel (~4.7 us/launch
+  // of pure launch overhead per step)
+  if (threadIdx.x == 0) {
+    const unsigned done = atomicAdd(done_counter, 1u) + 1u;
+    if (done == gridDim.x) {
+      *done_counter = 0u;
+      if (!grads_out) *t_dev = (int)t_new;
+    }
   }
 }
 
@@ -606,6 +665,8 @@ mlp_predict_gen_kernel(const float* __restrict__ X,  // [B][draw] raw fp32
   #pragma unroll
   for (int s = 0; s < G::SLOTS; ++s) acc1[s] = (f32x4){0.f, 0.f, 0.f, 0.f};
   const int nkt = (inp + G::TK - 1) / G::TK;
+  u16 (*Xs0)[G::XSS] = L.Xs(0);
+  u16 (*W1T0)[G::W1S] = L.W1T(0);
   for (int kt = 0; kt < nkt; ++kt) {
     const int k0 = kt * G::TK;
     for (int i = tid; i < RT * G::TK; i += G::BLOCK) {
@@ -615,9 +676,9 @@ mlp_predict_gen_kernel(const float* __restrict__ X,  // [B][draw] raw fp32
       if (row0 + r < B && c < draw) {
         v = (X[(row0 + r) * draw + c] - mean[c]) * invstd[c];
       }
-      L.Xs[r][j] = f2bf(v);
+      Xs0[r][j] = f2bf(v);
     }
-    load_w1t_tile(L, wimg, inp, k0, min(G::TK, inp - k0));
+    load_w1t_tile(L, 0, wimg, inp, k0, min(G::TK, inp - k0));
     __syncthreads();
     #pragma unroll
     for (int s = 0; s < G::SLOTS; ++s) {
@@ -628,8 +689,8 @@ mlp_predict_gen_kernel(const float* __restrict__ X,  // [B][draw] raw fp32
         f32x4 acc = acc1[s];
         #pragma unroll
         for (int ks = 0; ks < G::TK / 32; ++ks) {
-          const bf16x8 a = *(const bf16x8*)&L.W1T[mt * 16 + lr][ks * 32 + lg * 8];
-          const bf16x8 b = *(const bf16x8*)&L.Xs[rt * 16 + lr][ks * 32 + lg * 8];
+          const bf16x8 a = *(const bf16x8*)&W1T0[mt * 16 + lr][ks * 32 + lg * 8];
+          const bf16x8 b = *(const bf16x8*)&Xs0[rt * 16 + lr][ks * 32 + lg * 8];
           acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
         }
         acc1[s] = acc;
@@ -764,12 +825,9 @@ int rt_for_hid(int hid) {
 
 template <int HID, int RT>
 int launch_step_gen_t(const unsigned short* Xbf, const int* y, int B, int inp,
-                      int cls, unsigned short* wimg, float* master,
-                      unsigned short* bfmirror, float* m, float* v, int* t_dev,
+                      int cls, const unsigned short* wimg, const float* master,
                       float* slabs, int slab_stride, int max_slabs,
-                      unsigned* counter, float* loss_out, float invBtot, float lr,
-                      float beta1, float beta2, float eps, float* grads_out,
-                      hipStream_t stream) {
+                      float invBtot, hipStream_t stream) {
   using G = gen::GG<HID, RT>;
   const int blocks = (B + RT - 1) / RT;
   if (blocks > max_slabs) return -1;
@@ -784,8 +842,7 @@ int launch_step_gen_t(const unsigned short* Xbf, const int* y, int B, int inp,
   }
   hipLaunchKernelGGL((gen::mlp_step_gen_kernel<HID, RT>), dim3(blocks),
                      dim3(G::BLOCK), G::TOTAL, stream, Xbf, y, B, inp, cls, wimg,
-                     master, bfmirror, m, v, t_dev, slabs, slab_stride, counter,
-                     loss_out, invBtot, lr, beta1, beta2, eps, grads_out);
+                     master, slabs, slab_stride, invBtot);
   return 0;
 }
 
@@ -819,36 +876,41 @@ extern "C" {
 int gen_rt_for_hid(int hid) { return rt_for_hid(hid); }
 
 int launch_mlp_step_gen(const unsigned short* Xbf, const int* y, int B, int inp,
-                        int hid, int cls, unsigned short* wimg, float* master,
-                        unsigned short* bfmirror, float* m, float* v, int* t_dev,
-                        float* slabs, int slab_stride, int max_slabs,
-                        unsigned* counter, float* loss_out, float invBtot,
-                        float lr, float beta1, float beta2, float eps,
-                        float* grads_out, hipStream_t stream) {
+                        int hid, int cls, const unsigned short* wimg,
+                        const float* master, float* slabs, int slab_stride,
+                        int max_slabs, float invBtot, hipStream_t stream) {
   if (inp % 32 != 0 || cls > 16) return -3;
   switch (hid) {
     case 32:
-      return launch_step_gen_t<32, 128>(Xbf, y, B, inp, cls, wimg, master,
-                                        bfmirror, m, v, t_dev, slabs, slab_stride,
-                                        max_slabs, counter, loss_out, invBtot, lr,
-                                        beta1, beta2, eps, grads_out, stream);
+      return launch_step_gen_t<32, 128>(Xbf, y, B, inp, cls, wimg, master, slabs,
+                                        slab_stride, max_slabs, invBtot, stream);
     case 64:
-      return launch_step_gen_t<64, 128>(Xbf, y, B, inp, cls, wimg, master,
-                                        bfmirror, m, v, t_dev, slabs, slab_stride,
-                                        max_slabs, counter, loss_out, invBtot, lr,
-                                        beta1, beta2, eps, grads_out, stream);
+      return launch_step_gen_t<64, 128>(Xbf, y, B, inp, cls, wimg, master, slabs,
+                                        slab_stride, max_slabs, invBtot, stream);
     case 128:
-      return launch_step_gen_t<128, 64>(Xbf, y, B, inp, cls, wimg, master,
-                                        bfmirror, m, v, t_dev, slabs, slab_stride,
-                                        max_slabs, counter, loss_out, invBtot, lr,
-                                        beta1, beta2, eps, grads_out, stream);
+      return launch_step_gen_t<128, 64>(Xbf, y, B, inp, cls, wimg, master, slabs,
+                                        slab_stride, max_slabs, invBtot, stream);
     case 256:
-      return launch_step_gen_t<256, 32>(Xbf, y, B, inp, cls, wimg, master,
-                                        bfmirror, m, v, t_dev, slabs, slab_stride,
-                                        max_slabs, counter, loss_out, invBtot, lr,
-                                        beta1, beta2, eps, grads_out, stream);
+      return launch_step_gen_t<256, 32>(Xbf, y, B, inp, cls, wimg, master, slabs,
+                                        slab_stride, max_slabs, invBtot, stream);
   }
   return -3;
+}
+
+void launch_reduce_adam_gen(const float* slabs, int n_wg, int slab_stride,
+                            int inp, int hid, float* master,
+                            unsigned short* bfmirror, float* m, float* v,
+                            int* t_dev, float* loss_out, float lr, float beta1,
+                            float beta2, float eps, unsigned short* wimg,
+                            float* grads_out, unsigned* done_counter,
+                            hipStream_t stream) {
+  const int nparam = inp * hid + hid + hid * 16 + 16;
+  int blocks = (nparam + 255) / 256;
+  if (blocks > 512) blocks = 512;
+  hipLaunchKernelGGL(gen::reduce_adam_gen_kernel, dim3(blocks), dim3(256), 0,
+                     stream, slabs, n_wg, slab_stride, nparam, inp, hid, master,
+                     bfmirror, m, v, t_dev, loss_out, lr, beta1, beta2, eps,
+                     wimg, grads_out, done_counter);
 }
 
 int launch_mlp_predict_gen(const float* X, int B, int draw, int inp, int hid,

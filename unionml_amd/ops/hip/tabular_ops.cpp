@@ -13,9 +13,12 @@ void launch_standardize_apply(const float*, long long, int, int, const float*,
 // generalized-geometry kernels (tabular_gen.hip)
 int gen_rt_for_hid(int);
 int launch_mlp_step_gen(const unsigned short*, const int*, int, int, int, int,
-                        unsigned short*, float*, unsigned short*, float*, float*,
-                        int*, float*, int, int, unsigned*, float*, float, float,
-                        float, float, float, float*, hipStream_t);
+                        const unsigned short*, const float*, float*, int, int,
+                        float, hipStream_t);
+void launch_reduce_adam_gen(const float*, int, int, int, int, float*,
+                            unsigned short*, float*, float*, int*, float*,
+                            float, float, float, float, unsigned short*, float*,
+                            unsigned*, hipStream_t);
 int launch_mlp_predict_gen(const float*, int, int, int, int, int, const float*,
                            const float*, const unsigned short*, const float*,
                            int*, float*, hipStream_t);
@@ -218,20 +221,13 @@ void adam_step(torch::Tensor master, torch::Tensor bfmirror, torch::Tensor grads
 // ---------------------------------------------------------------------------
 
 bool mlp_step_gen(torch::Tensor Xbf, torch::Tensor y, int64_t hid, int64_t cls,
-                  torch::Tensor wimg, torch::Tensor master, torch::Tensor bfmirror,
-                  torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
-                  torch::Tensor slabs, torch::Tensor counter,
-                  torch::Tensor loss_out, double invBtot, double lr,
-                  double beta1, double beta2, double eps,
-                  c10::optional<torch::Tensor> grads_out = c10::nullopt) {
+                  torch::Tensor wimg, torch::Tensor master, torch::Tensor slabs,
+                  double invBtot) {
   check(Xbf, torch::kBFloat16, "Xbf");
   check(y, torch::kInt32, "y");
   check(wimg, torch::kBFloat16, "wimg");
   check(master, torch::kFloat32, "master");
-  check(bfmirror, torch::kBFloat16, "bfmirror");
   check(slabs, torch::kFloat32, "slabs");
-  check(counter, torch::kUInt32, "counter");
-  check(loss_out, torch::kFloat32, "loss_out");
   const int inp = (int)Xbf.size(1);
   const int nparam = inp * (int)hid + (int)hid + (int)hid * 16 + 16;
   TORCH_CHECK(inp % 32 == 0, "staged input width must be a multiple of 32");
@@ -240,24 +236,51 @@ bool mlp_step_gen(torch::Tensor Xbf, torch::Tensor y, int64_t hid, int64_t cls,
               "wimg too small for geometry");
   TORCH_CHECK(slabs.dim() == 2 && slabs.size(1) >= nparam + 2,
               "slabs must be [n][>= nparam+2]");
+  const int rc = launch_mlp_step_gen(
+      bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0), inp, (int)hid,
+      (int)cls, bf16_ptr(wimg), master.data_ptr<float>(),
+      slabs.data_ptr<float>(), (int)slabs.size(1), (int)slabs.size(0),
+      (float)invBtot, current_stream());
+  TORCH_CHECK(rc != -2, "mlp_step_gen: hipFuncSetAttribute(LDS) failed");
+  TORCH_CHECK(rc != -3, "mlp_step_gen: unsupported geometry (hid=", hid,
+              " inp=", inp, " cls=", cls, ")");
+  return rc == 0;  // false -> more WGs than slab rows, caller re-sizes
+}
+
+void reduce_adam_gen(torch::Tensor slabs, int64_t n_wg, int64_t inp, int64_t hid,
+                     torch::Tensor master, torch::Tensor bfmirror,
+                     torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
+                     torch::Tensor counter, torch::Tensor loss_out, double lr,
+                     double beta1, double beta2, double eps,
+                     c10::optional<torch::Tensor> wimg = c10::nullopt,
+                     c10::optional<torch::Tensor> grads_out = c10::nullopt) {
+  check(counter, torch::kUInt32, "counter");
+  check(slabs, torch::kFloat32, "slabs");
+  check(master, torch::kFloat32, "master");
+  check(bfmirror, torch::kBFloat16, "bfmirror");
+  check(loss_out, torch::kFloat32, "loss_out");
+  const int nparam = (int)(inp * hid + hid + hid * 16 + 16);
+  TORCH_CHECK(slabs.dim() == 2 && slabs.size(1) >= nparam + 2, "slab stride");
+  TORCH_CHECK(n_wg >= 1 && n_wg <= slabs.size(0), "n_wg out of range");
+  unsigned short* wimg_ptr = nullptr;
+  if (wimg.has_value()) {
+    check(*wimg, torch::kBFloat16, "wimg");
+    wimg_ptr = bf16_mut_ptr(*wimg);
+  }
   float* grads_ptr = nullptr;
   if (grads_out.has_value()) {
     check(*grads_out, torch::kFloat32, "grads_out");
     TORCH_CHECK(grads_out->numel() >= nparam + 1, "grads_out too small");
     grads_ptr = grads_out->data_ptr<float>();
   }
-  const int rc = launch_mlp_step_gen(
-      bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0), inp, (int)hid,
-      (int)cls, bf16_mut_ptr(wimg), master.data_ptr<float>(),
-      bf16_mut_ptr(bfmirror), m.data_ptr<float>(), v.data_ptr<float>(),
-      t_dev.data_ptr<int>(), slabs.data_ptr<float>(), (int)slabs.size(1),
-      (int)slabs.size(0), (unsigned*)counter.data_ptr(),
-      loss_out.data_ptr<float>(), (float)invBtot, (float)lr, (float)beta1,
-      (float)beta2, (float)eps, grads_ptr, current_stream());
-  TORCH_CHECK(rc != -2, "mlp_step_gen: hipFuncSetAttribute(LDS) failed");
-  TORCH_CHECK(rc != -3, "mlp_step_gen: unsupported geometry (hid=", hid,
-              " inp=", inp, " cls=", cls, ")");
-  return rc == 0;  // false -> more WGs than slab rows, caller re-sizes
+  launch_reduce_adam_gen(slabs.data_ptr<float>(), (int)n_wg,
+                         (int)slabs.size(1), (int)inp, (int)hid,
+                         master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
+                         m.data_ptr<float>(), v.data_ptr<float>(),
+                         t_dev.data_ptr<int>(), loss_out.data_ptr<float>(),
+                         (float)lr, (float)beta1, (float)beta2, (float)eps,
+                         wimg_ptr, grads_ptr, (unsigned*)counter.data_ptr(),
+                         current_stream());
 }
 
 void mlp_predict_gen(torch::Tensor X, int64_t inp, int64_t hid, int64_t cls,
@@ -328,13 +351,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gen_rt_for_hid", &gen_rt_for_hid,
         "rows-per-workgroup for a supported hidden width (0 = unsupported)");
   m.def("mlp_step_gen", &mlp_step_gen,
-        "generalized fused step: any (in,hid,cls) geometry; K-tiled MFMA "
-        "fwd+bwd + slab reduction + Adam in one launch (grads_out: reduce-only)",
+        "generalized fused step: any (in,hid,cls) geometry; double-buffered "
+        "K-tiled MFMA fwd+bwd producing per-WG gradient slabs",
         py::arg("Xbf"), py::arg("y"), py::arg("hid"), py::arg("cls"),
-        py::arg("wimg"), py::arg("master"), py::arg("bfmirror"), py::arg("m"),
-        py::arg("v"), py::arg("t_dev"), py::arg("slabs"), py::arg("counter"),
-        py::arg("loss_out"), py::arg("invBtot"), py::arg("lr"), py::arg("beta1"),
-        py::arg("beta2"), py::arg("eps"), py::arg("grads_out") = c10::nullopt);
+        py::arg("wimg"), py::arg("master"), py::arg("slabs"), py::arg("invBtot"));
+  m.def("reduce_adam_gen", &reduce_adam_gen,
+        "wide-grid slab reduction + fused Adam (grads_out: reduce-only, the "
+        "DP pre-collective mode)",
+        py::arg("slabs"), py::arg("n_wg"), py::arg("inp"), py::arg("hid"),
+        py::arg("master"), py::arg("bfmirror"), py::arg("m"), py::arg("v"),
+        py::arg("t_dev"), py::arg("counter"), py::arg("loss_out"),
+        py::arg("lr"), py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+        py::arg("wimg") = c10::nullopt, py::arg("grads_out") = c10::nullopt);
   m.def("mlp_predict_gen", &mlp_predict_gen,
         "generalized fused standardize+fwd+argmax",
         py::arg("X"), py::arg("inp"), py::arg("hid"), py::arg("cls"),

@@ -185,7 +185,9 @@ class TabularMLP:
         grads into self.grads (reduce-only mode: the DP pre-collective
         kernel — no Adam)."""
         g = self.g
-        self._ensure_slabs((Xbf.shape[0] + self._rows_per_wg() - 1) // self._rows_per_wg())
+        rpw = self._rows_per_wg()
+        n_wg = (Xbf.shape[0] + rpw - 1) // rpw
+        self._ensure_slabs(n_wg)
         loss_out = self.grads[g.nparam : g.nparam + 1]
         ext = hip_ext()
         if self.use_spec:
@@ -195,14 +197,19 @@ class TabularMLP:
                 invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
                 grads_out=self.grads, wimg=self.wimg,
             )
+            assert ok, "fused step slab capacity exceeded"
         else:
             ok = ext.mlp_step_gen(
-                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.bfmirror,
-                self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
-                invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
-                grads_out=self.grads,
+                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.slabs,
+                invBtot,
             )
-        assert ok, "fused step slab capacity exceeded"
+            assert ok, "fused step slab capacity exceeded"
+            ext.reduce_adam_gen(
+                self.slabs, n_wg, g.inp, g.hid, self.master, self.bfmirror,
+                self.m, self.v, self.t_dev, self.counter, loss_out,
+                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                wimg=self.wimg, grads_out=self.grads,
+            )
 
     def _adam(self, lr: float):
         g = self.g
@@ -241,9 +248,12 @@ class TabularMLP:
         return self.grads[g.nparam]
 
     def _fused_adam_step(self, Xbf, y, invBtot, lr, loss_out):
-        """One fully-fused optimizer step (fwd+bwd+reduce+Adam in one
-        launch for the specialized shape; one launch + image update for
-        generalized shapes)."""
+        """One optimizer step: a single fused launch for the specialized
+        shape; for generalized shapes the slab-producing step kernel
+        followed by the wide-grid reduce+Adam kernel (the kernel
+        boundary is the inter-WG barrier — full-chip bandwidth for the
+        large-nparam reduction instead of a handshake inside a tiny
+        grid)."""
         g = self.g
         ext = hip_ext()
         if self.use_spec:
@@ -252,13 +262,20 @@ class TabularMLP:
                 self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
                 invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
             )
+            assert ok, "fused step slab capacity exceeded"
         else:
+            rpw = self._rows_per_wg()
+            n_wg = (Xbf.shape[0] + rpw - 1) // rpw
             ok = ext.mlp_step_gen(
-                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.bfmirror,
-                self.m, self.v, self.t_dev, self.slabs, self.counter, loss_out,
-                invBtot, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.slabs,
+                invBtot,
             )
-        assert ok, "fused step slab capacity exceeded"
+            assert ok, "fused step slab capacity exceeded"
+            ext.reduce_adam_gen(
+                self.slabs, n_wg, g.inp, g.hid, self.master, self.bfmirror,
+                self.m, self.v, self.t_dev, self.counter, loss_out,
+                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
+            )
 
     def _train_epochs_fused(self, Xbf, y, batches, *, epochs, lr, use_graph) -> float:
         g = self.g
