@@ -876,24 +876,22 @@ extern "C" {
 int gen_rt_for_hid(int hid) { return rt_for_hid(hid); }
 
 int launch_mlp_step_gen(const unsigned short* Xbf, const int* y, int B, int inp,
-                        int hid, int cls, const unsigned short* wimg,
+                        int hid, int cls, int rt, const unsigned short* wimg,
                         const float* master, float* slabs, int slab_stride,
                         int max_slabs, float invBtot, hipStream_t stream) {
   if (inp % 32 != 0 || cls > 16) return -3;
-  switch (hid) {
-    case 32:
-      return launch_step_gen_t<32, 128>(Xbf, y, B, inp, cls, wimg, master, slabs,
-                                        slab_stride, max_slabs, invBtot, stream);
-    case 64:
-      return launch_step_gen_t<64, 128>(Xbf, y, B, inp, cls, wimg, master, slabs,
-                                        slab_stride, max_slabs, invBtot, stream);
-    case 128:
-      return launch_step_gen_t<128, 64>(Xbf, y, B, inp, cls, wimg, master, slabs,
-                                        slab_stride, max_slabs, invBtot, stream);
-    case 256:
-      return launch_step_gen_t<256, 32>(Xbf, y, B, inp, cls, wimg, master, slabs,
-                                        slab_stride, max_slabs, invBtot, stream);
-  }
+  #define STEP_CASE(H, R)                                                     \
+    if (hid == H && rt == R)                                                  \
+      return launch_step_gen_t<H, R>(Xbf, y, B, inp, cls, wimg, master, slabs,\
+                                     slab_stride, max_slabs, invBtot, stream);
+  STEP_CASE(32, 128)
+  STEP_CASE(32, 64)
+  STEP_CASE(64, 128)
+  STEP_CASE(64, 64)
+  STEP_CASE(128, 64)
+  STEP_CASE(128, 32)
+  STEP_CASE(256, 32)
+  #undef STEP_CASE
   return -3;
 }
 

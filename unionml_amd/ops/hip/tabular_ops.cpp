@@ -13,8 +13,8 @@ void launch_standardize_apply(const float*, long long, int, int, const float*,
 // generalized-geometry kernels (tabular_gen.hip)
 int gen_rt_for_hid(int);
 int launch_mlp_step_gen(const unsigned short*, const int*, int, int, int, int,
-                        const unsigned short*, const float*, float*, int, int,
-                        float, hipStream_t);
+                        int, const unsigned short*, const float*, float*, int,
+                        int, float, hipStream_t);
 void launch_reduce_adam_gen(const float*, int, int, int, int, float*,
                             unsigned short*, float*, float*, int*, float*,
                             float, float, float, float, unsigned short*, float*,
@@ -221,8 +221,8 @@ void adam_step(torch::Tensor master, torch::Tensor bfmirror, torch::Tensor grads
 // ---------------------------------------------------------------------------
 
 bool mlp_step_gen(torch::Tensor Xbf, torch::Tensor y, int64_t hid, int64_t cls,
-                  torch::Tensor wimg, torch::Tensor master, torch::Tensor slabs,
-                  double invBtot) {
+                  int64_t rt, torch::Tensor wimg, torch::Tensor master,
+                  torch::Tensor slabs, double invBtot) {
   check(Xbf, torch::kBFloat16, "Xbf");
   check(y, torch::kInt32, "y");
   check(wimg, torch::kBFloat16, "wimg");
@@ -238,12 +238,12 @@ bool mlp_step_gen(torch::Tensor Xbf, torch::Tensor y, int64_t hid, int64_t cls,
               "slabs must be [n][>= nparam+2]");
   const int rc = launch_mlp_step_gen(
       bf16_ptr(Xbf), y.data_ptr<int>(), (int)Xbf.size(0), inp, (int)hid,
-      (int)cls, bf16_ptr(wimg), master.data_ptr<float>(),
+      (int)cls, (int)rt, bf16_ptr(wimg), master.data_ptr<float>(),
       slabs.data_ptr<float>(), (int)slabs.size(1), (int)slabs.size(0),
       (float)invBtot, current_stream());
   TORCH_CHECK(rc != -2, "mlp_step_gen: hipFuncSetAttribute(LDS) failed");
   TORCH_CHECK(rc != -3, "mlp_step_gen: unsupported geometry (hid=", hid,
-              " inp=", inp, " cls=", cls, ")");
+              " inp=", inp, " cls=", cls, " rt=", rt, ")");
   return rc == 0;  // false -> more WGs than slab rows, caller re-sizes
 }
 
@@ -354,7 +354,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "generalized fused step: any (in,hid,cls) geometry; double-buffered "
         "K-tiled MFMA fwd+bwd producing per-WG gradient slabs",
         py::arg("Xbf"), py::arg("y"), py::arg("hid"), py::arg("cls"),
-        py::arg("wimg"), py::arg("master"), py::arg("slabs"), py::arg("invBtot"));
+        py::arg("rt"), py::arg("wimg"), py::arg("master"), py::arg("slabs"),
+        py::arg("invBtot"));
   m.def("reduce_adam_gen", &reduce_adam_gen,
         "wide-grid slab reduction + fused Adam (grads_out: reduce-only, the "
         "DP pre-collective mode)",

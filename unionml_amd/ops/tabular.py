@@ -104,10 +104,19 @@ class TabularMLP:
     def _wimg_len(self) -> int:
         return 4224 if self.use_spec else self.g.wimg_n
 
-    def _rows_per_wg(self) -> int:
+    #: per hidden width: (default rows/WG, small-batch rows/WG). The
+    #: small variant keeps >= ~16 workgroups in flight at modest batch
+    #: sizes — the fwd/bwd phase is bandwidth-bound and achievable
+    #: bandwidth scales with resident CUs.
+    _GEN_RTS = {32: (128, 64), 64: (128, 64), 128: (64, 32), 256: (32, 32)}
+
+    def _rows_per_wg(self, batch: Optional[int] = None) -> int:
         if self.use_spec:
             return 128
-        return {32: 128, 64: 128, 128: 64, 256: 32}[self.g.hid]
+        big, small = self._GEN_RTS[self.g.hid]
+        if batch is not None and (batch + big - 1) // big < 16:
+            return small
+        return big
 
     def _build_wimg(self):
         if self.wimg is None:
@@ -185,7 +194,7 @@ class TabularMLP:
         grads into self.grads (reduce-only mode: the DP pre-collective
         kernel — no Adam)."""
         g = self.g
-        rpw = self._rows_per_wg()
+        rpw = self._rows_per_wg(Xbf.shape[0])
         n_wg = (Xbf.shape[0] + rpw - 1) // rpw
         self._ensure_slabs(n_wg)
         loss_out = self.grads[g.nparam : g.nparam + 1]
@@ -200,8 +209,8 @@ class TabularMLP:
             assert ok, "fused step slab capacity exceeded"
         else:
             ok = ext.mlp_step_gen(
-                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.slabs,
-                invBtot,
+                Xbf, y, g.hid, g.classes, rpw, self.wimg, self.master,
+                self.slabs, invBtot,
             )
             assert ok, "fused step slab capacity exceeded"
             ext.reduce_adam_gen(
@@ -264,11 +273,12 @@ class TabularMLP:
             )
             assert ok, "fused step slab capacity exceeded"
         else:
-            rpw = self._rows_per_wg()
+            rpw = self._rows_per_wg(Xbf.shape[0])
             n_wg = (Xbf.shape[0] + rpw - 1) // rpw
+            self._ensure_slabs(n_wg)
             ok = ext.mlp_step_gen(
-                Xbf, y, g.hid, g.classes, self.wimg, self.master, self.slabs,
-                invBtot,
+                Xbf, y, g.hid, g.classes, rpw, self.wimg, self.master,
+                self.slabs, invBtot,
             )
             assert ok, "fused step slab capacity exceeded"
             ext.reduce_adam_gen(
@@ -279,8 +289,12 @@ class TabularMLP:
 
     def _train_epochs_fused(self, Xbf, y, batches, *, epochs, lr, use_graph) -> float:
         g = self.g
-        rpw = self._rows_per_wg()
-        self._ensure_slabs(max((bs + rpw - 1) // rpw for _, bs in batches))
+        self._ensure_slabs(
+            max(
+                (bs + self._rows_per_wg(bs) - 1) // self._rows_per_wg(bs)
+                for _, bs in batches
+            )
+        )
         loss_out = self.grads[g.nparam : g.nparam + 1]
 
         def run_epoch():
