@@ -178,3 +178,106 @@ def test_reader_retries_kwarg_passthrough():
     assert task.retries == 2
     out = task(n=4)
     assert len(out) == 4 and len(attempts) == 2
+
+
+# ------------------- workflow DAG hardening (r02) ---------------------
+
+
+def test_workflow_rejects_unknown_input_binding():
+    from unionml_amd.task import Task, Workflow, WorkflowError
+
+    wf = Workflow("w", inputs=["a"], outputs=[("out", ("node", 0, None))])
+    t = Task(lambda x: x, "t")
+    with pytest.raises(WorkflowError, match="unknown input"):
+        wf.add_node(t, {"x": ("input", "nope")})
+
+
+def test_workflow_rejects_forward_node_reference():
+    from unionml_amd.task import Task, Workflow, WorkflowError
+
+    wf = Workflow("w", inputs=["a"], outputs=[("out", ("node", 0, None))])
+    t = Task(lambda x: x, "t")
+    with pytest.raises(WorkflowError, match="acyclic"):
+        wf.add_node(t, {"x": ("node", 3, None)})  # node 3 doesn't exist
+
+
+def test_workflow_validates_outputs():
+    from unionml_amd.task import Task, Workflow, WorkflowError
+
+    wf = Workflow("w", inputs=["a"], outputs=[("out", ("node", 9, None))])
+    wf.add_node(Task(lambda x: x, "t"), {"x": ("input", "a")})
+    with pytest.raises(WorkflowError):
+        wf(a=1)
+
+
+def test_workflow_bad_output_index_is_clear_error():
+    from unionml_amd.task import Task, Workflow, WorkflowError
+
+    wf = Workflow("w", inputs=["a"], outputs=[("out", ("node", 0, 5))])
+    wf.add_node(Task(lambda x: (x,), "t"), {"x": ("input", "a")})
+    with pytest.raises(WorkflowError, match="output index"):
+        wf(a=1)
+
+
+def test_workflow_typed_inputs():
+    from unionml_amd.task import Task, Workflow
+
+    wf = Workflow("w", inputs={"n": int}, outputs=[("out", ("node", 0, None))])
+    wf.add_node(Task(lambda n: n * 2, "t"), {"n": ("input", "n")})
+    assert wf(n=4) == 8
+    with pytest.raises(TypeError, match="expects"):
+        wf(n="four")
+
+
+def test_workflow_parallel_branches():
+    """Independent nodes in one dependency wave run concurrently when
+    parallel=True: two 0.2s sleepers finish in well under 0.4s."""
+    import time as _time
+
+    from unionml_amd.task import Task, Workflow
+
+    def slow(tag):
+        def fn(x):
+            _time.sleep(0.2)
+            return f"{tag}:{x}"
+
+        return fn
+
+    wf = Workflow(
+        "w",
+        inputs=["a"],
+        outputs=[("out", ("node", 2, None))],
+        parallel=True,
+    )
+    n0 = wf.add_node(Task(slow("l"), "left"), {"x": ("input", "a")})
+    n1 = wf.add_node(Task(slow("r"), "right"), {"x": ("input", "a")})
+    wf.add_node(
+        Task(lambda l, r: (l, r), "join"),
+        {"l": ("node", n0, None), "r": ("node", n1, None)},
+    )
+    t0 = _time.perf_counter()
+    out = wf(a=1)
+    wall = _time.perf_counter() - t0
+    assert out == ("l:1", "r:1")
+    assert wall < 0.38, f"branches did not run concurrently ({wall:.2f}s)"
+
+    # same graph, sequential executor: takes >= 0.4s
+    wf.parallel = False
+    t0 = _time.perf_counter()
+    wf(a=1)
+    assert _time.perf_counter() - t0 >= 0.39
+
+
+def test_workflow_dependency_order_not_insertion_order():
+    """A node added early but depending on a later wave's result is
+    impossible by construction; conversely, waves compute correct
+    depths for diamond graphs."""
+    from unionml_amd.task import Task, Workflow
+
+    wf = Workflow("diamond", inputs=["a"], outputs=[("out", ("node", 3, None))])
+    top = wf.add_node(Task(lambda x: x + 1, "top"), {"x": ("input", "a")})
+    l = wf.add_node(Task(lambda x: x * 2, "l"), {"x": ("node", top, None)})
+    r = wf.add_node(Task(lambda x: x * 3, "r"), {"x": ("node", top, None)})
+    wf.add_node(Task(lambda p, q: p + q, "join"), {"p": ("node", l, None), "q": ("node", r, None)})
+    assert wf(a=1) == 2 * 2 + 2 * 3
+    assert [sorted(w) for w in wf._waves()] == [[0], [1, 2], [3]]

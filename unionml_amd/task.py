@@ -172,24 +172,96 @@ class WorkflowNode:
     kwargs_from: Optional[str] = None
 
 
-class Workflow:
-    """A static DAG of tasks with named inputs/outputs, executed in-process.
+class WorkflowError(ValueError):
+    """Raised for malformed workflow DAGs (bad bindings, cycles,
+    unknown inputs) — at graph-build time, not execution time."""
 
-    Nodes execute in insertion order (the builder adds them already
-    topologically sorted); each node's inputs are bound either to
-    workflow inputs or to upstream node outputs.
+
+class Workflow:
+    """A static DAG of tasks with named (optionally typed) inputs and
+    outputs.
+
+    Every binding is validated when the node is added: "input" sources
+    must name a declared workflow input, "node" sources must reference
+    an already-added node (so the graph is acyclic by construction, and
+    a bad output index fails before execution, not mid-run). Execution
+    is dependency-driven: nodes are grouped into ready waves, and a
+    wave with multiple independent nodes runs its branches concurrently
+    when the workflow is built with ``parallel=True``.
+
+    ``inputs`` may be a sequence of names, or a mapping name -> type
+    for runtime input type checking.
     """
 
-    def __init__(self, name: str, inputs: Sequence[str], outputs: Sequence[Tuple[str, Tuple]]):
+    def __init__(
+        self,
+        name: str,
+        inputs,
+        outputs: Sequence[Tuple[str, Tuple]],
+        parallel: bool = False,
+    ):
         self.name = name
-        self.input_names = list(inputs)
-        # outputs: list of (output_name, source) with source like bindings
+        if isinstance(inputs, dict):
+            self.input_names = list(inputs)
+            self.input_types: Dict[str, Any] = dict(inputs)
+        else:
+            self.input_names = list(inputs)
+            self.input_types = {}
         self.outputs = list(outputs)
+        self.parallel = parallel
         self.nodes: List[WorkflowNode] = []
 
+    def _check_source(self, source: Tuple, n_nodes: int, where: str) -> None:
+        if not isinstance(source, tuple) or not source:
+            raise WorkflowError(f"workflow '{self.name}': malformed binding {source!r} in {where}")
+        kind = source[0]
+        if kind == "input":
+            if source[1] not in self.input_names:
+                raise WorkflowError(
+                    f"workflow '{self.name}': {where} references unknown input "
+                    f"{source[1]!r} (declared: {self.input_names})"
+                )
+        elif kind == "node":
+            _, node_idx, _out_idx = source
+            if not (0 <= node_idx < n_nodes):
+                raise WorkflowError(
+                    f"workflow '{self.name}': {where} references node {node_idx}, but "
+                    f"only {n_nodes} node(s) exist upstream — bindings may only point "
+                    "at already-added nodes (this keeps the DAG acyclic by construction)"
+                )
+        else:
+            raise WorkflowError(f"workflow '{self.name}': unknown binding kind {kind!r} in {where}")
+
     def add_node(self, task: Task, bindings: Dict[str, Tuple], kwargs_from: Optional[str] = None) -> int:
+        for pname, src in bindings.items():
+            self._check_source(src, len(self.nodes), f"node '{task.name}' param '{pname}'")
+        if kwargs_from is not None and kwargs_from not in self.input_names:
+            raise WorkflowError(
+                f"workflow '{self.name}': node '{task.name}' kwargs_from references "
+                f"unknown input {kwargs_from!r}"
+            )
         self.nodes.append(WorkflowNode(task=task, bindings=bindings, kwargs_from=kwargs_from))
         return len(self.nodes) - 1
+
+    def validate(self) -> None:
+        """Validate the finished graph (outputs may reference any node)."""
+        for out_name, src in self.outputs:
+            self._check_source(src, len(self.nodes), f"output '{out_name}'")
+
+    def _deps(self, node: WorkflowNode) -> set:
+        return {src[1] for src in node.bindings.values() if src[0] == "node"}
+
+    def _waves(self) -> List[List[int]]:
+        """Group node indices into dependency waves (every node's deps
+        are in strictly earlier waves)."""
+        depth: Dict[int, int] = {}
+        for i, node in enumerate(self.nodes):
+            deps = self._deps(node)
+            depth[i] = (max(depth[d] for d in deps) + 1) if deps else 0
+        waves: Dict[int, List[int]] = {}
+        for i, d in depth.items():
+            waves.setdefault(d, []).append(i)
+        return [waves[d] for d in sorted(waves)]
 
     def _resolve(self, source: Tuple, wf_inputs: Dict[str, Any], node_results: List[Any]):
         kind = source[0]
@@ -198,27 +270,68 @@ class Workflow:
         if kind == "node":
             _, node_idx, out_idx = source
             result = node_results[node_idx]
-            return result if out_idx is None else result[out_idx]
-        raise ValueError(f"unknown binding source {source!r}")
+            if out_idx is not None:
+                try:
+                    return result[out_idx]
+                except (IndexError, KeyError, TypeError) as exc:
+                    raise WorkflowError(
+                        f"workflow '{self.name}': node {node_idx} "
+                        f"({self.nodes[node_idx].task.name}) returned "
+                        f"{type(result).__name__} without output index {out_idx!r}"
+                    ) from exc
+            return result
+        raise WorkflowError(f"unknown binding source {source!r}")
+
+    def _check_input_types(self, wf_inputs: Dict[str, Any]) -> None:
+        for name, expected in self.input_types.items():
+            if expected is None or expected is Any or name not in wf_inputs:
+                continue
+            value = wf_inputs[name]
+            if value is None:
+                continue
+            origin = getattr(expected, "__origin__", expected)
+            if isinstance(origin, type) and not isinstance(value, origin):
+                raise TypeError(
+                    f"workflow '{self.name}' input '{name}' expects "
+                    f"{expected}, got {type(value).__name__}"
+                )
+
+    def _run_node(self, node: WorkflowNode, wf_inputs, node_results):
+        kwargs = {
+            pname: self._resolve(src, wf_inputs, node_results)
+            for pname, src in node.bindings.items()
+        }
+        if node.kwargs_from is not None:
+            extra = wf_inputs.get(node.kwargs_from) or {}
+            kwargs.update(extra)
+        t0 = time.perf_counter()
+        out = node.task(**kwargs)
+        logger.debug(
+            "workflow %s: task %s took %.3fs", self.name, node.task.name, time.perf_counter() - t0
+        )
+        return out
 
     def __call__(self, **wf_inputs):
         missing = set(self.input_names) - set(wf_inputs)
         if missing:
             raise TypeError(f"workflow '{self.name}' missing inputs: {sorted(missing)}")
-        node_results: List[Any] = []
-        for node in self.nodes:
-            kwargs = {
-                pname: self._resolve(src, wf_inputs, node_results)
-                for pname, src in node.bindings.items()
-            }
-            if node.kwargs_from is not None:
-                extra = wf_inputs.get(node.kwargs_from) or {}
-                kwargs.update(extra)
-            t0 = time.perf_counter()
-            node_results.append(node.task(**kwargs))
-            logger.debug(
-                "workflow %s: task %s took %.3fs", self.name, node.task.name, time.perf_counter() - t0
-            )
+        self.validate()
+        self._check_input_types(wf_inputs)
+        node_results: List[Any] = [None] * len(self.nodes)
+        for wave in self._waves():
+            if self.parallel and len(wave) > 1:
+                import concurrent.futures as cf
+
+                with cf.ThreadPoolExecutor(max_workers=len(wave)) as pool:
+                    futures = {
+                        i: pool.submit(self._run_node, self.nodes[i], wf_inputs, node_results)
+                        for i in wave
+                    }
+                    for i, fut in futures.items():
+                        node_results[i] = fut.result()
+            else:
+                for i in wave:
+                    node_results[i] = self._run_node(self.nodes[i], wf_inputs, node_results)
         outs = tuple(self._resolve(src, wf_inputs, node_results) for _, src in self.outputs)
         return outs[0] if len(outs) == 1 else outs
 
