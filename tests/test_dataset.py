@@ -240,3 +240,42 @@ def test_feature_loader_fast_path_preserves_dtypes():
     assert feats["b"].dtype.kind == "f"
     assert feats["c"].dtype.kind == "b"
     assert feats["a"].tolist() == [1, 2]
+
+
+def test_feature_type_union_with_custom_loader():
+    """A custom feature_loader with a different return type makes
+    feature_type a FeatureTypeUnion[dataset_type, loaded_type]
+    (reference dataset.py:405-424); bentoml IO inference unwraps the
+    serve-time arm."""
+    import typing
+
+    from unionml_amd.dataset import FeatureTypeUnion
+
+    ds = Dataset(name="ftu", features=["a"], targets=["t"])
+
+    @ds.reader
+    def reader() -> pd.DataFrame:
+        return pd.DataFrame({"a": [1.0], "t": [0]})
+
+    assert ds.feature_type in (pd.DataFrame, typing.Any)
+
+    @ds.feature_loader
+    def feature_loader(data) -> np.ndarray:
+        return np.asarray(data)
+
+    ft = ds.feature_type
+    assert typing.get_origin(ft) is FeatureTypeUnion
+    assert typing.get_args(ft)[1] is np.ndarray
+
+    from unionml_amd.services.bentoml import infer_io_descriptor
+
+    assert infer_io_descriptor(ft) == "NumpyNdarray"
+
+    # predictor guards accept either arm of the union
+    from unionml_amd import type_guards as tg
+
+    assert tg._types_compatible(ft, np.ndarray)
+    strict = FeatureTypeUnion[pd.DataFrame, np.ndarray]
+    assert tg._types_compatible(strict, np.ndarray)
+    assert tg._types_compatible(strict, pd.DataFrame)
+    assert not tg._types_compatible(strict, int)

@@ -31,9 +31,19 @@ from unionml_amd.task import Task, inner_task
 from unionml_amd.tracker import TrackedInstance
 
 
-class FeatureTypeUnion:
-    """Marker generic for 'raw-or-loaded features' unions
-    (reference: dataset.py:30-31)."""
+import typing as _typing
+
+DT = _typing.TypeVar("DT")  # dataset (parser) feature type
+FT = _typing.TypeVar("FT")  # serve-time loaded feature type
+
+
+class FeatureTypeUnion(_typing.Generic[DT, FT]):
+    """'raw-or-loaded features' union marker (reference dataset.py:30-31):
+    when a custom feature_loader returns a different type than the
+    parser's feature element, :attr:`Dataset.feature_type` is
+    ``FeatureTypeUnion[dataset_type, loaded_type]`` — consumers that
+    need the serve-time type (bentoml IO inference) unwrap the second
+    argument."""
 
 
 def _train_test_split(data: pd.DataFrame, test_size: float, shuffle: bool, random_state: int):
@@ -216,9 +226,20 @@ class Dataset(TrackedInstance):
 
     @property
     def feature_type(self) -> Type:
-        """The type of one split's feature element (reference:
-        dataset.py:412-424)."""
-        return self.parser_return_types[self._parser_feature_key]
+        """The type of one split's feature element; with a custom
+        feature_loader producing a different type, the
+        :class:`FeatureTypeUnion` of both (reference: dataset.py:405-424)."""
+        dataset_type = self.parser_return_types[self._parser_feature_key]
+        if self._feature_loader == self._default_feature_loader:
+            return dataset_type
+        loaded_type = (
+            signature(self._feature_loader).return_annotation
+            if self._feature_transformer == self._default_feature_transformer
+            else signature(self._feature_transformer).return_annotation
+        )
+        if loaded_type is inspect.Signature.empty or loaded_type == dataset_type:
+            return dataset_type
+        return FeatureTypeUnion[dataset_type, loaded_type]  # type: ignore[misc]
 
     # ------------------------------------------------------------------
     # compiled task

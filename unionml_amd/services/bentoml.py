@@ -52,6 +52,8 @@ def infer_io_descriptor(type_: Type) -> Optional[str]:
     name (reference IO_DESCRIPTOR_MAPPING: services/bentoml.py:33-38)."""
     import typing
 
+    if getattr(typing.get_origin(type_), "__name__", "") == "FeatureTypeUnion":
+        type_ = typing.get_args(type_)[1]  # serve-time (loaded) type
     origin = typing.get_origin(type_) or type_
     try:
         import numpy as np

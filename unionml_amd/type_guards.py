@@ -58,6 +58,9 @@ def _types_compatible(expected: Any, actual: Any) -> bool:
         return True
     if expected is Any or actual is Any:
         return True
+    # a FeatureTypeUnion[dataset_type, loaded_type] accepts either arm
+    if getattr(typing.get_origin(expected), "__name__", "") == "FeatureTypeUnion":
+        return any(_types_compatible(arm, actual) for arm in typing.get_args(expected))
     exp_origin = typing.get_origin(expected) or expected
     act_origin = typing.get_origin(actual) or actual
     if exp_origin is typing.Union or act_origin is typing.Union:
