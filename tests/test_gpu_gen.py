@@ -16,6 +16,8 @@ SHAPES = [
     (64, 32, 12),     # gen kernels at the small geometry (spec shape is 10-class)
     (100, 50, 7),     # pads every axis: inp 128, hid 64, classes 7
     (33, 250, 3),     # extreme padding: inp 64, hid 256 (RT=32 instantiation)
+    (64, 32, 26),     # two-tile classifier head (cpad 32)
+    (200, 128, 32),   # full 32-class head
 ]
 
 
@@ -86,7 +88,7 @@ def test_gen_step_grads_vs_reference(ext, dev, shape):
     assert (W2g[:, g.classes :] == 0).all(), "class-pad cols leaked gradient"
 
 
-@pytest.mark.parametrize("shape", [(784, 128, 10), (100, 50, 7)],
+@pytest.mark.parametrize("shape", [(784, 128, 10), (100, 50, 7), (64, 32, 26)],
                          ids=lambda s: "x".join(map(str, s)))
 def test_gen_reduce_mode_matches_fused(ext, dev, shape):
     """grads_out mode (the DP pre-collective step) + adam_step_gen must

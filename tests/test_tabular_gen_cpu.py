@@ -32,11 +32,21 @@ def test_geometry_padding_math():
 
 def test_geometry_rejects_unsupported():
     with pytest.raises(ValueError, match="classes"):
-        Geometry(64, 32, 17)
+        Geometry(64, 32, 33)
     with pytest.raises(ValueError, match="hidden"):
         Geometry(64, 300, 10)
     with pytest.raises(ValueError):
         Geometry(0, 32, 10)
+
+
+def test_geometry_wide_class_head():
+    """17..32 classes use the two-tile classifier head (cpad 32)."""
+    g = Geometry(64, 32, 26)
+    assert g.cpad == 32
+    assert g.off_b2 == g.off_w2 + 32 * 32
+    assert g.nparam == 64 * 32 + 32 + 32 * 32 + 32
+    assert Geometry(64, 32, 16).cpad == 16
+    assert Geometry(64, 32, 17).cpad == 32
 
 
 def test_parametric_matches_legacy_at_digits_shape():
@@ -130,3 +140,22 @@ def test_legacy_pickle_without_geometry_loads():
     assert clone.g.is_specialized
     X = torch.randn(8, 64)
     assert clone.predict(X).shape == (8,)
+
+
+@pytest.mark.parametrize("shape", [(64, 32, 26), (100, 60, 20)],
+                         ids=lambda s: "x".join(map(str, s)))
+def test_cpu_train_predict_wide_classes(shape):
+    inf, hid, cls = shape
+    clf = TabularMLP(in_features=inf, hidden=hid, classes=cls, device="cpu", seed=0)
+    assert clf.g.cpad == 32
+    g = torch.Generator().manual_seed(2)
+    n = 512
+    centers = torch.randn(cls, inf, generator=g) * 3.0
+    y = torch.randint(0, cls, (n,), generator=g, dtype=torch.int32)
+    X = centers[y.long()] + torch.randn(n, inf, generator=g) * 0.5
+    clf.fit_standardizer(X)
+    loss = clf.train_epochs(clf.stage(X), y, epochs=10, batch_size=128, lr=5e-3)
+    assert loss == loss
+    preds = clf.predict(X)
+    acc = (preds == y).float().mean().item()
+    assert acc > 0.9, f"{shape}: accuracy {acc}, loss {loss}"

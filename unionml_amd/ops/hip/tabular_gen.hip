@@ -60,13 +60,15 @@ __device__ __forceinline__ float fast_rcp(float x) {
 constexpr int A16(int x) { return (x + 15) & ~15; }
 constexpr int IMAX(int a, int b) { return a > b ? a : b; }
 
-// Geometry: LDS layout + tile bookkeeping, all from <HID, RT>.
-template <int HID_, int RT_>
+// Geometry: LDS layout + tile bookkeeping, all from <HID, RT, CP>.
+template <int HID_, int RT_, int CP_ = 16>
 struct GG {
   static constexpr int HID = HID_;
   static constexpr int RT = RT_;        // batch rows per workgroup
   static constexpr int TK = 64;         // k-tile over the input width
-  static constexpr int CPAD = 16;       // classes padded to one MFMA tile
+  static constexpr int CPAD = CP_;      // classes padded to MFMA tiles (16/32)
+  static constexpr int NCT = CPAD / 16; // class tiles in the fwd2 head
+  static_assert(CPAD == 16 || CPAD == 32, "classifier head: 16 or 32 classes");
   static constexpr int WAVES = 8;
   static constexpr int BLOCK = WAVES * 64;
 
@@ -184,7 +186,7 @@ __device__ __forceinline__ int wimg_w2t_off(int hid, int inp) {
 }
 
 // Adam-phase image maintenance: param i -> packed-image stores
-template <int HID>
+template <int HID, int CP>
 __device__ __forceinline__ void wimg_write_gen(u16* __restrict__ wimg, int inp,
                                                int off_b1, int off_w2, int off_b2,
                                                int i, u16 wb) {
@@ -192,10 +194,10 @@ __device__ __forceinline__ void wimg_write_gen(u16* __restrict__ wimg, int inp,
     const int in = i / HID;
     const int h = i % HID;
     wimg[h * inp + in] = wb;
-  } else if (i >= off_w2 && i < off_b2) { // W2: j = h*16 + c
+  } else if (i >= off_w2 && i < off_b2) { // W2: j = h*CP + c
     const int j = i - off_w2;
-    const int h = j >> 4;
-    const int c = j & 15;
+    const int h = j / CP;
+    const int c = j % CP;
     wimg[wimg_w2s_off(HID, inp) + h * 32 + c] = wb;
     wimg[wimg_w2t_off(HID, inp) + c * HID + h] = wb;
   }
@@ -275,9 +277,12 @@ __device__ __forceinline__ void load_w2_images(const G& L, const u16* __restrict
 
 template <typename G>
 __device__ __forceinline__ void zero_dl_pad(const G& L) {
-  for (int i = threadIdx.x; i < G::RT * G::CPAD; i += G::BLOCK) {
-    const int r = i / G::CPAD, c = G::CPAD + (i % G::CPAD);
-    L.DLs[r][c] = 0;
+  if constexpr (G::CPAD < 32) {
+    constexpr int PADW = 32 - G::CPAD;
+    for (int i = threadIdx.x; i < G::RT * PADW; i += G::BLOCK) {
+      const int r = i / PADW, c = G::CPAD + (i % PADW);
+      L.DLs[r][c] = 0;
+    }
   }
 }
 
@@ -285,7 +290,7 @@ __device__ __forceinline__ void zero_dl_pad(const G& L) {
 // fused generalized step kernel
 // ---------------------------------------------------------------------------
 
-template <int HID, int RT>
+template <int HID, int RT, int CP>
 __global__ void __launch_bounds__(512)
 mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero-padded)
                     const int* __restrict__ y, int B,
@@ -294,7 +299,7 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
                     const float* __restrict__ master,  // biases prefetch
                     float* __restrict__ slabs, int slab_stride,
                     float invBtot) {
-  using G = GG<HID, RT>;
+  using G = GG<HID, RT, CP>;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   G L;
   L.carve(smem);
@@ -386,61 +391,86 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
   }
   __syncthreads();
 
-  // ---- fwd2: L^T = W2T @ B(Hs); wave-shuffle softmax + xent ----------------
+  // ---- fwd2: L^T = W2T @ B(Hs); wave-shuffle softmax + xent over NCT
+  // class tiles (classes <= 16: one MFMA tile; 17..32: two tiles, the
+  // max/sum folds across the register-resident tile axis first, then
+  // the 4 lane-groups) ------------------------------------------------------
   for (int t2 = wave; t2 < RT / 16; t2 += G::WAVES) {
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    f32x4 acc[G::NCT];
     #pragma unroll
-    for (int ks = 0; ks < HID / 32; ++ks) {
-      const bf16x8 a = *(const bf16x8*)&L.W2Tt[lr_][ks * 32 + lg * 8];
-      const bf16x8 b = *(const bf16x8*)&L.Hs[t2 * 16 + lr_][ks * 32 + lg * 8];
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    for (int ct = 0; ct < G::NCT; ++ct) {
+      acc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int ks = 0; ks < HID / 32; ++ks) {
+        const bf16x8 a = *(const bf16x8*)&L.W2Tt[ct * 16 + lr_][ks * 32 + lg * 8];
+        const bf16x8 b = *(const bf16x8*)&L.Hs[t2 * 16 + lr_][ks * 32 + lg * 8];
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[ct], 0, 0, 0);
+      }
     }
     const int row = t2 * 16 + lr_;
     const bool valid = (row0 + row) < (long long)B;
     const int label = valid ? y[row0 + row] : -1;
 
-    float logit[4];
+    float logit[G::NCT][4];
     float mx = -1e30f;
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int c = lg * 4 + r;
-      logit[r] = (c < cls) ? acc[r] + L.b2s[c] : -1e30f;
-      mx = fmaxf(mx, logit[r]);
+    for (int ct = 0; ct < G::NCT; ++ct) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int c = ct * 16 + lg * 4 + r;
+        logit[ct][r] = (c < cls) ? acc[ct][r] + L.b2s[c] : -1e30f;
+        mx = fmaxf(mx, logit[ct][r]);
+      }
     }
     mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
     mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
-    float e[4], ssum = 0.f;
+    float e[G::NCT][4], ssum = 0.f;
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int c = lg * 4 + r;
-      e[r] = (c < cls) ? __expf(logit[r] - mx) : 0.f;
-      ssum += e[r];
+    for (int ct = 0; ct < G::NCT; ++ct) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int c = ct * 16 + lg * 4 + r;
+        e[ct][r] = (c < cls) ? __expf(logit[ct][r] - mx) : 0.f;
+        ssum += e[ct][r];
+      }
     }
     ssum += __shfl_xor(ssum, 16, 64);
     ssum += __shfl_xor(ssum, 32, 64);
     const float rs = fast_rcp(ssum);
     const float logs = __logf(ssum);
-    float db2_acc[4];
+    float db2_acc[G::NCT][4];
     float loss_acc = 0.f;
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int c = lg * 4 + r;
-      const float dl = valid ? (e[r] * rs - (c == label ? 1.f : 0.f)) * invBtot : 0.f;
-      const u16 dlb = f2bf(dl);
-      L.DLs[row][c] = dlb;
-      L.DLT[c][row] = dlb;
-      db2_acc[r] = dl;
-      if (valid && c == label) loss_acc = -(logit[r] - mx - logs) * invBtot;
+    for (int ct = 0; ct < G::NCT; ++ct) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int c = ct * 16 + lg * 4 + r;
+        const float dl =
+            valid ? (e[ct][r] * rs - (c == label ? 1.f : 0.f)) * invBtot : 0.f;
+        const u16 dlb = f2bf(dl);
+        L.DLs[row][c] = dlb;
+        L.DLT[c][row] = dlb;
+        db2_acc[ct][r] = dl;
+        if (valid && c == label) loss_acc = -(logit[ct][r] - mx - logs) * invBtot;
+      }
     }
     #pragma unroll
     for (int bit = 1; bit < 16; bit <<= 1) {
       #pragma unroll
-      for (int r = 0; r < 4; ++r) db2_acc[r] += __shfl_xor(db2_acc[r], bit, 64);
+      for (int ct = 0; ct < G::NCT; ++ct) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) db2_acc[ct][r] += __shfl_xor(db2_acc[ct][r], bit, 64);
+      }
       loss_acc += __shfl_xor(loss_acc, bit, 64);
     }
     if (lr_ == 0) {
       #pragma unroll
-      for (int r = 0; r < 4; ++r) atomicAdd(&L.db2[lg * 4 + r], db2_acc[r]);
+      for (int ct = 0; ct < G::NCT; ++ct) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          atomicAdd(&L.db2[ct * 16 + lg * 4 + r], db2_acc[ct][r]);
+        }
+      }
     }
     loss_acc += __shfl_xor(loss_acc, 16, 64);
     loss_acc += __shfl_xor(loss_acc, 32, 64);
@@ -479,18 +509,20 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
   float* slab = slabs + (long long)blockIdx.x * slab_stride;
 
   // ---- dW2 = HT @ B(DLT) -> slab (plain stores) ----------------------------
-  for (int t = wave; t < HID / 16; t += G::WAVES) {
+  for (int t = wave; t < (HID / 16) * G::NCT; t += G::WAVES) {
+    const int ht = t % (HID / 16);
+    const int ct = t / (HID / 16);
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     #pragma unroll
     for (int ks = 0; ks < RT / 32; ++ks) {
-      const bf16x8 a = *(const bf16x8*)&L.HT[t * 16 + lr_][ks * 32 + lg * 8];
-      const bf16x8 b = *(const bf16x8*)&L.DLT[lr_][ks * 32 + lg * 8];
+      const bf16x8 a = *(const bf16x8*)&L.HT[ht * 16 + lr_][ks * 32 + lg * 8];
+      const bf16x8 b = *(const bf16x8*)&L.DLT[ct * 16 + lr_][ks * 32 + lg * 8];
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
     }
     #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int h = t * 16 + lg * 4 + r;
-      slab[off_w2 + h * G::CPAD + lr_] = acc[r];
+      const int h = ht * 16 + lg * 4 + r;
+      slab[off_w2 + h * G::CPAD + ct * 16 + lr_] = acc[r];
     }
   }
 
@@ -560,7 +592,7 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
 
 __global__ void __launch_bounds__(256)
 reduce_adam_gen_kernel(const float* __restrict__ slabs, int n_wg,
-                       int slab_stride, int nparam, int inp, int hid,
+                       int slab_stride, int nparam, int inp, int hid, int cpad,
                        float* __restrict__ master, u16* __restrict__ bfmirror,
                        float* __restrict__ m, float* __restrict__ v,
                        int* __restrict__ t_dev,
@@ -574,7 +606,7 @@ reduce_adam_gen_kernel(const float* __restrict__ slabs, int n_wg,
   const float corr2 = fast_rcp(1.f - __powf(beta2, t_new));
   const int off_b1 = inp * hid;
   const int off_w2 = off_b1 + hid;
-  const int off_b2 = off_w2 + hid * 16;
+  const int off_b2 = off_w2 + hid * cpad;
   for (int i = blockIdx.x * 256 + threadIdx.x; i <= nparam;
        i += gridDim.x * 256) {
     float g0 = 0.f, g1 = 0.f, g2 = 0.f, g3 = 0.f;
@@ -608,7 +640,7 @@ reduce_adam_gen_kernel(const float* __restrict__ slabs, int n_wg,
         wimg[(i % hid) * inp + (i / hid)] = wb;
       } else if (i >= off_w2 && i < off_b2) {
         const int j = i - off_w2;
-        const int h = j >> 4, c = j & 15;
+        const int h = j / cpad, c = j % cpad;
         wimg[hid * inp + h * 32 + c] = wb;
         wimg[hid * inp + hid * 32 + c * hid + h] = wb;
       }
@@ -631,7 +663,7 @@ reduce_adam_gen_kernel(const float* __restrict__ slabs, int n_wg,
 // generalized fused predict: standardize + fwd + argmax (+softmax)
 // ---------------------------------------------------------------------------
 
-template <int HID, int RT>
+template <int HID, int RT, int CP>
 __global__ void __launch_bounds__(512)
 mlp_predict_gen_kernel(const float* __restrict__ X,  // [B][draw] raw fp32
                        int B, int draw, int inp, int cls,
@@ -641,7 +673,7 @@ mlp_predict_gen_kernel(const float* __restrict__ X,  // [B][draw] raw fp32
                        const float* __restrict__ master,  // biases
                        int* __restrict__ preds,
                        float* __restrict__ probs /* optional [B][cls] */) {
-  using G = GG<HID, RT>;
+  using G = GG<HID, RT, CP>;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   G L;
   L.carve(smem);
@@ -715,24 +747,32 @@ mlp_predict_gen_kernel(const float* __restrict__ X,  // [B][draw] raw fp32
   }
   __syncthreads();
 
-  // fwd2 + argmax / softmax over the class axis
+  // fwd2 + argmax / softmax over NCT class tiles
   for (int t2 = wave; t2 < RT / 16; t2 += G::WAVES) {
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    f32x4 acc[G::NCT];
     #pragma unroll
-    for (int ks = 0; ks < HID / 32; ++ks) {
-      const bf16x8 a = *(const bf16x8*)&L.W2Tt[lr][ks * 32 + lg * 8];
-      const bf16x8 b = *(const bf16x8*)&L.Hs[t2 * 16 + lr][ks * 32 + lg * 8];
-      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    for (int ct = 0; ct < G::NCT; ++ct) {
+      acc[ct] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      #pragma unroll
+      for (int ks = 0; ks < HID / 32; ++ks) {
+        const bf16x8 a = *(const bf16x8*)&L.W2Tt[ct * 16 + lr][ks * 32 + lg * 8];
+        const bf16x8 b = *(const bf16x8*)&L.Hs[t2 * 16 + lr][ks * 32 + lg * 8];
+        acc[ct] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[ct], 0, 0, 0);
+      }
     }
     const int row = t2 * 16 + lr;
     float best = -1e30f;
     int bcol = cls;
-    float logit[4];
+    float logit[G::NCT][4];
     #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int c = lg * 4 + r;
-      logit[r] = (c < cls) ? acc[r] + L.b2s[c] : -1e30f;
-      if (logit[r] > best) { best = logit[r]; bcol = c; }  // ties: lowest class
+    for (int ct = 0; ct < G::NCT; ++ct) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int c = ct * 16 + lg * 4 + r;
+        logit[ct][r] = (c < cls) ? acc[ct][r] + L.b2s[c] : -1e30f;
+        // ties pick the lowest class (match torch.argmax)
+        if (logit[ct][r] > best) { best = logit[ct][r]; bcol = c; }
+      }
     }
     #pragma unroll
     for (int d = 16; d < 64; d <<= 1) {
@@ -742,20 +782,26 @@ mlp_predict_gen_kernel(const float* __restrict__ X,  // [B][draw] raw fp32
     }
     if (lg == 0 && row0 + row < B) preds[row0 + row] = bcol;
     if (probs != nullptr) {
-      float ssum = 0.f, e[4];
+      float ssum = 0.f, e[G::NCT][4];
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int c = lg * 4 + r;
-        e[r] = (c < cls) ? __expf(logit[r] - best) : 0.f;
-        ssum += e[r];
+      for (int ct = 0; ct < G::NCT; ++ct) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int c = ct * 16 + lg * 4 + r;
+          e[ct][r] = (c < cls) ? __expf(logit[ct][r] - best) : 0.f;
+          ssum += e[ct][r];
+        }
       }
       ssum += __shfl_xor(ssum, 16, 64);
       ssum += __shfl_xor(ssum, 32, 64);
       const float rs = fast_rcp(ssum);
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int c = lg * 4 + r;
-        if (c < cls && row0 + row < B) probs[(row0 + row) * cls + c] = e[r] * rs;
+      for (int ct = 0; ct < G::NCT; ++ct) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int c = ct * 16 + lg * 4 + r;
+          if (c < cls && row0 + row < B) probs[(row0 + row) * cls + c] = e[ct][r] * rs;
+        }
       }
     }
   }
@@ -771,7 +817,7 @@ __global__ void __launch_bounds__(256)
 adam_step_gen_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
                      const float* __restrict__ grads, float* __restrict__ m,
                      float* __restrict__ v, const int* __restrict__ t_dev,
-                     int nparam, int inp, int hid,
+                     int nparam, int inp, int hid, int cpad,
                      float lr, float beta1, float beta2, float eps,
                      u16* __restrict__ wimg) {
   const float t = (float)(*t_dev + 1);
@@ -779,7 +825,7 @@ adam_step_gen_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
   const float corr2 = fast_rcp(1.f - __powf(beta2, t));
   const int off_b1 = inp * hid;
   const int off_w2 = off_b1 + hid;
-  const int off_b2 = off_w2 + hid * 16;
+  const int off_b2 = off_w2 + hid * cpad;
   for (int i = blockIdx.x * 256 + threadIdx.x; i < nparam; i += gridDim.x * 256) {
     const float g = grads[i];
     const float mi = beta1 * m[i] + (1.f - beta1) * g;
@@ -795,7 +841,7 @@ adam_step_gen_kernel(float* __restrict__ master, u16* __restrict__ bfmirror,
         wimg[(i % hid) * inp + (i / hid)] = wb;
       } else if (i >= off_w2 && i < off_b2) {
         const int j = i - off_w2;
-        const int h = j >> 4, c = j & 15;
+        const int h = j / cpad, c = j % cpad;
         wimg[hid * inp + h * 32 + c] = wb;
         wimg[hid * inp + hid * 32 + c * hid + h] = wb;
       }
@@ -823,46 +869,46 @@ int rt_for_hid(int hid) {
   return 0;
 }
 
-template <int HID, int RT>
+template <int HID, int RT, int CP>
 int launch_step_gen_t(const unsigned short* Xbf, const int* y, int B, int inp,
                       int cls, const unsigned short* wimg, const float* master,
                       float* slabs, int slab_stride, int max_slabs,
                       float invBtot, hipStream_t stream) {
-  using G = gen::GG<HID, RT>;
+  using G = gen::GG<HID, RT, CP>;
   const int blocks = (B + RT - 1) / RT;
   if (blocks > max_slabs) return -1;
   static int done = 0;
   if (!done) {
-    if (hipFuncSetAttribute((const void*)gen::mlp_step_gen_kernel<HID, RT>,
+    if (hipFuncSetAttribute((const void*)gen::mlp_step_gen_kernel<HID, RT, CP>,
                             hipFuncAttributeMaxDynamicSharedMemorySize,
                             G::TOTAL) != hipSuccess) {
       return -2;
     }
     done = 1;
   }
-  hipLaunchKernelGGL((gen::mlp_step_gen_kernel<HID, RT>), dim3(blocks),
+  hipLaunchKernelGGL((gen::mlp_step_gen_kernel<HID, RT, CP>), dim3(blocks),
                      dim3(G::BLOCK), G::TOTAL, stream, Xbf, y, B, inp, cls, wimg,
                      master, slabs, slab_stride, invBtot);
   return 0;
 }
 
-template <int HID, int RT>
+template <int HID, int RT, int CP>
 int launch_predict_gen_t(const float* X, int B, int draw, int inp, int cls,
                          const float* mean, const float* invstd,
                          const unsigned short* wimg, const float* master,
                          int* preds, float* probs, hipStream_t stream) {
-  using G = gen::GG<HID, RT>;
+  using G = gen::GG<HID, RT, CP>;
   static int done = 0;
   if (!done) {
-    if (hipFuncSetAttribute((const void*)gen::mlp_predict_gen_kernel<HID, RT>,
-                            hipFuncAttributeMaxDynamicSharedMemorySize,
-                            G::TOTAL) != hipSuccess) {
+    if (hipFuncSetAttribute(
+            (const void*)gen::mlp_predict_gen_kernel<HID, RT, CP>,
+            hipFuncAttributeMaxDynamicSharedMemorySize, G::TOTAL) != hipSuccess) {
       return -2;
     }
     done = 1;
   }
   const int blocks = (B + RT - 1) / RT;
-  hipLaunchKernelGGL((gen::mlp_predict_gen_kernel<HID, RT>), dim3(blocks),
+  hipLaunchKernelGGL((gen::mlp_predict_gen_kernel<HID, RT, CP>), dim3(blocks),
                      dim3(G::BLOCK), G::TOTAL, stream, X, B, draw, inp, cls, mean,
                      invstd, wimg, master, preds, probs);
   return 0;
@@ -879,70 +925,74 @@ int launch_mlp_step_gen(const unsigned short* Xbf, const int* y, int B, int inp,
                         int hid, int cls, int rt, const unsigned short* wimg,
                         const float* master, float* slabs, int slab_stride,
                         int max_slabs, float invBtot, hipStream_t stream) {
-  if (inp % 32 != 0 || cls > 16) return -3;
-  #define STEP_CASE(H, R)                                                     \
-    if (hid == H && rt == R)                                                  \
-      return launch_step_gen_t<H, R>(Xbf, y, B, inp, cls, wimg, master, slabs,\
-                                     slab_stride, max_slabs, invBtot, stream);
-  STEP_CASE(32, 128)
-  STEP_CASE(32, 64)
-  STEP_CASE(64, 128)
-  STEP_CASE(64, 64)
-  STEP_CASE(128, 64)
-  STEP_CASE(128, 32)
-  STEP_CASE(256, 32)
+  if (inp % 32 != 0 || cls > 32) return -3;
+  #define STEP_CASE(H, R, C)                                                  \
+    if (hid == H && rt == R && cls <= C && (C == 16 ? cls <= 16 : cls > 16))  \
+      return launch_step_gen_t<H, R, C>(Xbf, y, B, inp, cls, wimg, master,    \
+                                        slabs, slab_stride, max_slabs,        \
+                                        invBtot, stream);
+  STEP_CASE(32, 128, 16)
+  STEP_CASE(32, 64, 16)
+  STEP_CASE(64, 128, 16)
+  STEP_CASE(64, 64, 16)
+  STEP_CASE(128, 64, 16)
+  STEP_CASE(128, 32, 16)
+  STEP_CASE(256, 32, 16)
+  STEP_CASE(32, 128, 32)
+  STEP_CASE(64, 128, 32)
+  STEP_CASE(128, 64, 32)
+  STEP_CASE(256, 32, 32)
   #undef STEP_CASE
   return -3;
 }
 
 void launch_reduce_adam_gen(const float* slabs, int n_wg, int slab_stride,
-                            int inp, int hid, float* master,
+                            int inp, int hid, int cpad, float* master,
                             unsigned short* bfmirror, float* m, float* v,
                             int* t_dev, float* loss_out, float lr, float beta1,
                             float beta2, float eps, unsigned short* wimg,
                             float* grads_out, unsigned* done_counter,
                             hipStream_t stream) {
-  const int nparam = inp * hid + hid + hid * 16 + 16;
+  const int nparam = inp * hid + hid + hid * cpad + cpad;
   int blocks = (nparam + 255) / 256;
   if (blocks > 512) blocks = 512;
   hipLaunchKernelGGL(gen::reduce_adam_gen_kernel, dim3(blocks), dim3(256), 0,
-                     stream, slabs, n_wg, slab_stride, nparam, inp, hid, master,
-                     bfmirror, m, v, t_dev, loss_out, lr, beta1, beta2, eps,
-                     wimg, grads_out, done_counter);
+                     stream, slabs, n_wg, slab_stride, nparam, inp, hid, cpad,
+                     master, bfmirror, m, v, t_dev, loss_out, lr, beta1, beta2,
+                     eps, wimg, grads_out, done_counter);
 }
 
 int launch_mlp_predict_gen(const float* X, int B, int draw, int inp, int hid,
                            int cls, const float* mean, const float* invstd,
                            const unsigned short* wimg, const float* master,
                            int* preds, float* probs, hipStream_t stream) {
-  if (inp % 32 != 0 || cls > 16) return -3;
-  switch (hid) {
-    case 32:
-      return launch_predict_gen_t<32, 128>(X, B, draw, inp, cls, mean, invstd,
+  if (inp % 32 != 0 || cls > 32) return -3;
+  #define PRED_CASE(H, R, C)                                                  \
+    if (hid == H && (C == 16 ? cls <= 16 : cls > 16))                         \
+      return launch_predict_gen_t<H, R, C>(X, B, draw, inp, cls, mean, invstd,\
                                            wimg, master, preds, probs, stream);
-    case 64:
-      return launch_predict_gen_t<64, 128>(X, B, draw, inp, cls, mean, invstd,
-                                           wimg, master, preds, probs, stream);
-    case 128:
-      return launch_predict_gen_t<128, 64>(X, B, draw, inp, cls, mean, invstd,
-                                           wimg, master, preds, probs, stream);
-    case 256:
-      return launch_predict_gen_t<256, 32>(X, B, draw, inp, cls, mean, invstd,
-                                           wimg, master, preds, probs, stream);
-  }
+  PRED_CASE(32, 128, 16)
+  PRED_CASE(64, 128, 16)
+  PRED_CASE(128, 64, 16)
+  PRED_CASE(256, 32, 16)
+  PRED_CASE(32, 128, 32)
+  PRED_CASE(64, 128, 32)
+  PRED_CASE(128, 64, 32)
+  PRED_CASE(256, 32, 32)
+  #undef PRED_CASE
   return -3;
 }
 
 void launch_adam_step_gen(float* master, unsigned short* bfmirror,
                           const float* grads, float* m, float* v, int* t_dev,
-                          int nparam, int inp, int hid, float lr, float beta1,
-                          float beta2, float eps, unsigned short* wimg,
-                          hipStream_t stream) {
+                          int nparam, int inp, int hid, int cpad, float lr,
+                          float beta1, float beta2, float eps,
+                          unsigned short* wimg, hipStream_t stream) {
   int blocks = (nparam + 255) / 256;
   if (blocks > 512) blocks = 512;
   hipLaunchKernelGGL(gen::adam_step_gen_kernel, dim3(blocks), dim3(256), 0,
                      stream, master, bfmirror, grads, m, v, t_dev, nparam, inp,
-                     hid, lr, beta1, beta2, eps, wimg);
+                     hid, cpad, lr, beta1, beta2, eps, wimg);
   hipLaunchKernelGGL(gen::bump_t_kernel, dim3(1), dim3(1), 0, stream, t_dev);
 }
 

@@ -15,7 +15,7 @@ int gen_rt_for_hid(int);
 int launch_mlp_step_gen(const unsigned short*, const int*, int, int, int, int,
                         int, const unsigned short*, const float*, float*, int,
                         int, float, hipStream_t);
-void launch_reduce_adam_gen(const float*, int, int, int, int, float*,
+void launch_reduce_adam_gen(const float*, int, int, int, int, int, float*,
                             unsigned short*, float*, float*, int*, float*,
                             float, float, float, float, unsigned short*, float*,
                             unsigned*, hipStream_t);
@@ -23,7 +23,7 @@ int launch_mlp_predict_gen(const float*, int, int, int, int, int, const float*,
                            const float*, const unsigned short*, const float*,
                            int*, float*, hipStream_t);
 void launch_adam_step_gen(float*, unsigned short*, const float*, float*, float*,
-                          int*, int, int, int, float, float, float, float,
+                          int*, int, int, int, int, float, float, float, float,
                           unsigned short*, hipStream_t);
 void launch_mlp_step(const unsigned short*, const int*, int, const unsigned short*,
                      const unsigned short*, const float*, float*, float,
@@ -229,10 +229,11 @@ bool mlp_step_gen(torch::Tensor Xbf, torch::Tensor y, int64_t hid, int64_t cls,
   check(master, torch::kFloat32, "master");
   check(slabs, torch::kFloat32, "slabs");
   const int inp = (int)Xbf.size(1);
-  const int nparam = inp * (int)hid + (int)hid + (int)hid * 16 + 16;
+  const int cpad = cls <= 16 ? 16 : 32;
+  const int nparam = (int)(inp * hid + hid + hid * cpad + cpad);
   TORCH_CHECK(inp % 32 == 0, "staged input width must be a multiple of 32");
   TORCH_CHECK(master.numel() >= nparam, "master too small for geometry");
-  TORCH_CHECK(wimg.numel() >= (int64_t)hid * inp + hid * 32 + 16 * hid,
+  TORCH_CHECK(wimg.numel() >= (int64_t)hid * inp + hid * 32 + cpad * hid,
               "wimg too small for geometry");
   TORCH_CHECK(slabs.dim() == 2 && slabs.size(1) >= nparam + 2,
               "slabs must be [n][>= nparam+2]");
@@ -248,7 +249,7 @@ bool mlp_step_gen(torch::Tensor Xbf, torch::Tensor y, int64_t hid, int64_t cls,
 }
 
 void reduce_adam_gen(torch::Tensor slabs, int64_t n_wg, int64_t inp, int64_t hid,
-                     torch::Tensor master, torch::Tensor bfmirror,
+                     int64_t cpad, torch::Tensor master, torch::Tensor bfmirror,
                      torch::Tensor m, torch::Tensor v, torch::Tensor t_dev,
                      torch::Tensor counter, torch::Tensor loss_out, double lr,
                      double beta1, double beta2, double eps,
@@ -259,7 +260,8 @@ void reduce_adam_gen(torch::Tensor slabs, int64_t n_wg, int64_t inp, int64_t hid
   check(master, torch::kFloat32, "master");
   check(bfmirror, torch::kBFloat16, "bfmirror");
   check(loss_out, torch::kFloat32, "loss_out");
-  const int nparam = (int)(inp * hid + hid + hid * 16 + 16);
+  TORCH_CHECK(cpad == 16 || cpad == 32, "cpad must be 16 or 32");
+  const int nparam = (int)(inp * hid + hid + hid * cpad + cpad);
   TORCH_CHECK(slabs.dim() == 2 && slabs.size(1) >= nparam + 2, "slab stride");
   TORCH_CHECK(n_wg >= 1 && n_wg <= slabs.size(0), "n_wg out of range");
   unsigned short* wimg_ptr = nullptr;
@@ -274,7 +276,7 @@ void reduce_adam_gen(torch::Tensor slabs, int64_t n_wg, int64_t inp, int64_t hid
     grads_ptr = grads_out->data_ptr<float>();
   }
   launch_reduce_adam_gen(slabs.data_ptr<float>(), (int)n_wg,
-                         (int)slabs.size(1), (int)inp, (int)hid,
+                         (int)slabs.size(1), (int)inp, (int)hid, (int)cpad,
                          master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
                          m.data_ptr<float>(), v.data_ptr<float>(),
                          t_dev.data_ptr<int>(), loss_out.data_ptr<float>(),
@@ -308,13 +310,14 @@ void mlp_predict_gen(torch::Tensor X, int64_t inp, int64_t hid, int64_t cls,
 
 void adam_step_gen(torch::Tensor master, torch::Tensor bfmirror,
                    torch::Tensor grads, torch::Tensor m, torch::Tensor v,
-                   torch::Tensor t_dev, int64_t inp, int64_t hid, double lr,
-                   double beta1, double beta2, double eps,
+                   torch::Tensor t_dev, int64_t inp, int64_t hid, int64_t cpad,
+                   double lr, double beta1, double beta2, double eps,
                    c10::optional<torch::Tensor> wimg = c10::nullopt) {
   check(master, torch::kFloat32, "master");
   check(bfmirror, torch::kBFloat16, "bfmirror");
   check(grads, torch::kFloat32, "grads");
-  const int nparam = (int)(inp * hid + hid + hid * 16 + 16);
+  TORCH_CHECK(cpad == 16 || cpad == 32, "cpad must be 16 or 32");
+  const int nparam = (int)(inp * hid + hid + hid * cpad + cpad);
   TORCH_CHECK(master.numel() >= nparam, "master too small for geometry");
   unsigned short* wimg_ptr = nullptr;
   if (wimg.has_value()) {
@@ -324,8 +327,8 @@ void adam_step_gen(torch::Tensor master, torch::Tensor bfmirror,
   launch_adam_step_gen(master.data_ptr<float>(), bf16_mut_ptr(bfmirror),
                        grads.data_ptr<float>(), m.data_ptr<float>(),
                        v.data_ptr<float>(), t_dev.data_ptr<int>(), nparam,
-                       (int)inp, (int)hid, (float)lr, (float)beta1, (float)beta2,
-                       (float)eps, wimg_ptr, current_stream());
+                       (int)inp, (int)hid, (int)cpad, (float)lr, (float)beta1,
+                       (float)beta2, (float)eps, wimg_ptr, current_stream());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -360,8 +363,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "wide-grid slab reduction + fused Adam (grads_out: reduce-only, the "
         "DP pre-collective mode)",
         py::arg("slabs"), py::arg("n_wg"), py::arg("inp"), py::arg("hid"),
-        py::arg("master"), py::arg("bfmirror"), py::arg("m"), py::arg("v"),
-        py::arg("t_dev"), py::arg("counter"), py::arg("loss_out"),
+        py::arg("cpad"), py::arg("master"), py::arg("bfmirror"), py::arg("m"),
+        py::arg("v"), py::arg("t_dev"), py::arg("counter"), py::arg("loss_out"),
         py::arg("lr"), py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
         py::arg("wimg") = c10::nullopt, py::arg("grads_out") = c10::nullopt);
   m.def("mlp_predict_gen", &mlp_predict_gen,
@@ -373,6 +376,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "generalized fused Adam (runtime param count; DP path)",
         py::arg("master"), py::arg("bfmirror"), py::arg("grads"), py::arg("m"),
         py::arg("v"), py::arg("t_dev"), py::arg("inp"), py::arg("hid"),
-        py::arg("lr"), py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
-        py::arg("wimg") = c10::nullopt);
+        py::arg("cpad"), py::arg("lr"), py::arg("beta1"), py::arg("beta2"),
+        py::arg("eps"), py::arg("wimg") = c10::nullopt);
 }

@@ -16,7 +16,7 @@ NPARAM = 2608
 
 # hidden widths with a compiled gen-kernel instantiation (tabular_gen.hip)
 SUPPORTED_HID = (32, 64, 128, 256)
-MAX_CLASSES = 16
+MAX_CLASSES = 32
 
 
 class Geometry:
@@ -36,8 +36,8 @@ class Geometry:
             raise ValueError(f"bad geometry ({in_features}, {hidden}, {classes})")
         if classes > MAX_CLASSES:
             raise ValueError(
-                f"classes={classes} unsupported: the fused classifier head is one "
-                f"MFMA tile ({MAX_CLASSES} classes max)"
+                f"classes={classes} unsupported: the fused classifier head spans "
+                f"at most two MFMA tiles ({MAX_CLASSES} classes max)"
             )
         if hidden > SUPPORTED_HID[-1]:
             raise ValueError(
@@ -48,14 +48,14 @@ class Geometry:
         self.classes = classes
         self.inp = (in_features + 31) // 32 * 32
         self.hid = next(h for h in SUPPORTED_HID if h >= hidden)
-        self.cpad = CPAD
+        self.cpad = 16 if classes <= 16 else 32
         self.off_w1 = 0
         self.off_b1 = self.inp * self.hid
         self.off_w2 = self.off_b1 + self.hid
         self.off_b2 = self.off_w2 + self.hid * self.cpad
         self.nparam = self.off_b2 + self.cpad
         self.slab_stride = (self.nparam + 2 + 15) // 16 * 16
-        # packed weight images: W1T [hid][inp] + W2s [hid][32] + W2T [16][hid]
+        # packed weight images: W1T [hid][inp] + W2s [hid][32] + W2T [cpad][hid]
         self.wimg_n = self.hid * self.inp + self.hid * 32 + self.cpad * self.hid
         # the hand-tuned specialized kernel path covers exactly this shape
         self.is_specialized = (in_features, hidden, classes) == (64, 32, 10)

@@ -114,6 +114,8 @@ class TabularMLP:
         if self.use_spec:
             return 128
         big, small = self._GEN_RTS[self.g.hid]
+        if self.g.cpad > 16:
+            return big  # 32-class head compiles only at the default rows/WG
         if batch is not None and (batch + big - 1) // big < 16:
             return small
         return big
@@ -134,8 +136,8 @@ class TabularMLP:
             o1 = g.hid * g.inp
             o2 = o1 + g.hid * 32
             img[:o1].view(g.hid, g.inp).copy_(W1.t().bfloat16())           # W1T[h][k]
-            img[o1:o2].view(g.hid, 32)[:, :CPAD] = W2.bfloat16()           # W2s K-pad
-            img[o2:].view(CPAD, g.hid).copy_(W2.t().bfloat16())            # W2T[c][h]
+            img[o1:o2].view(g.hid, 32)[:, : g.cpad] = W2.bfloat16()        # W2s K-pad
+            img[o2:].view(g.cpad, g.hid).copy_(W2.t().bfloat16())          # W2T[c][h]
         self.wimg.copy_(img.to(self.device))
 
     def _ensure_slabs(self, n_wg: int):
@@ -214,9 +216,9 @@ class TabularMLP:
             )
             assert ok, "fused step slab capacity exceeded"
             ext.reduce_adam_gen(
-                self.slabs, n_wg, g.inp, g.hid, self.master, self.bfmirror,
-                self.m, self.v, self.t_dev, self.counter, loss_out,
-                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                self.slabs, n_wg, g.inp, g.hid, g.cpad, self.master,
+                self.bfmirror, self.m, self.v, self.t_dev, self.counter,
+                loss_out, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
                 wimg=self.wimg, grads_out=self.grads,
             )
 
@@ -231,7 +233,8 @@ class TabularMLP:
         else:
             ext.adam_step_gen(
                 self.master, self.bfmirror, self.grads, self.m, self.v, self.t_dev,
-                g.inp, g.hid, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
+                g.inp, g.hid, g.cpad, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS,
+                wimg=self.wimg,
             )
 
     def _step(self, Xbf: torch.Tensor, y: torch.Tensor, invBtot: float, lr: float,
@@ -282,9 +285,9 @@ class TabularMLP:
             )
             assert ok, "fused step slab capacity exceeded"
             ext.reduce_adam_gen(
-                self.slabs, n_wg, g.inp, g.hid, self.master, self.bfmirror,
-                self.m, self.v, self.t_dev, self.counter, loss_out,
-                lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
+                self.slabs, n_wg, g.inp, g.hid, g.cpad, self.master,
+                self.bfmirror, self.m, self.v, self.t_dev, self.counter,
+                loss_out, lr, ADAM_BETA1, ADAM_BETA2, ADAM_EPS, wimg=self.wimg,
             )
 
     def _train_epochs_fused(self, Xbf, y, batches, *, epochs, lr, use_graph) -> float:
