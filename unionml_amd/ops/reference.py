@@ -25,8 +25,8 @@ class Geometry:
 
     The HIP kernels run on a zero-padded image of the model: input width
     padded to a multiple of 32 (MFMA K granularity), hidden padded up to
-    a compiled instantiation width, classes padded to one 16-wide MFMA
-    tile. Zero padding is mathematically exact under Adam-from-zero-init
+    a compiled instantiation width, classes padded to one or two 16-wide
+    MFMA tiles. Zero padding is mathematically exact under Adam-from-zero-init
     (zero init + zero gradient -> zero update), so the padded model's
     real sub-block evolves identically to the unpadded math.
     """
