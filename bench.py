@@ -331,11 +331,32 @@ def validate_graph_capture(clf, eager_step, dist, device):
     restore_state(clf, s0)
     torch.cuda.synchronize()
 
-    ok = True
+    g = None
     try:
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             eager_step(0)
+    except RuntimeError as exc:
+        print(f"[bench] graph capture validation failed ({exc})", file=sys.stderr)
+        g = None
+
+    # agree on capture success BEFORE anyone replays: a rank replaying a
+    # captured RCCL collective whose peers never replay would hang the
+    # job (capture failures are not guaranteed to be symmetric)
+    if dist is not None:
+        flag = torch.tensor([0.0 if g is None else 1.0], device=device)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+        if flag.item() < 0.5:
+            restore_state(clf, s0)
+            torch.cuda.synchronize()
+            return False
+    elif g is None:
+        restore_state(clf, s0)
+        torch.cuda.synchronize()
+        return False
+
+    ok = True
+    try:
         g.replay()
         torch.cuda.synchronize()
         got = checksum_state(clf)
@@ -347,10 +368,10 @@ def validate_graph_capture(clf, eager_step, dist, device):
         if not ok:
             print(f"[bench] graph validation mismatch: eager={want} replay={got}",
                   file=sys.stderr)
-        del g
     except RuntimeError as exc:
-        print(f"[bench] graph capture validation failed ({exc})", file=sys.stderr)
+        print(f"[bench] graph replay validation failed ({exc})", file=sys.stderr)
         ok = False
+    del g
     restore_state(clf, s0)
     torch.cuda.synchronize()
 
