@@ -84,7 +84,11 @@ class MiniAsgiLambdaAdapter:
     # -- invocation ---------------------------------------------------
 
     def __call__(self, event: Dict[str, Any], context: Any = None) -> Dict[str, Any]:
-        return asyncio.new_event_loop().run_until_complete(self._handle(event))
+        loop = asyncio.new_event_loop()
+        try:
+            return loop.run_until_complete(self._handle(event))
+        finally:
+            loop.close()
 
     async def _handle(self, event: Dict[str, Any]) -> Dict[str, Any]:
         if not self._lifespan_started:
