@@ -83,27 +83,34 @@ def evaluator(net: DPNet, features: pd.DataFrame, target: pd.DataFrame) -> float
 N_ROWS = 256
 
 
-def _make_data():
+def _make_data(shape=(64, 32, 10)):
     torch.manual_seed(7)
-    X = torch.randn(N_ROWS, 64) * 1.5
-    y = torch.randint(0, 10, (N_ROWS,), dtype=torch.int32)
+    inf, _, cls = shape
+    X = torch.randn(N_ROWS, inf) * 1.5
+    y = torch.randint(0, cls, (N_ROWS,), dtype=torch.int32)
     return X, y
 
 
-def _tabular_worker(rank, world, port, out_dir):
+def _make_clf(shape):
+    from unionml_amd.ops.tabular import TabularMLP
+
+    inf, hid, cls = shape
+    return TabularMLP(in_features=inf, hidden=hid, classes=cls, device="cpu", seed=1)
+
+
+def _tabular_worker(rank, world, port, out_dir, shape):
     import torch.distributed as dist
 
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
     try:
-        from unionml_amd.ops.tabular import TabularMLP
         from unionml_amd.parallel.ddp import shard
 
-        X, y = _make_data()
+        X, y = _make_data(shape)
         Xr, yr = shard(X, rank, world), shard(y, rank, world)
-        clf = TabularMLP(device="cpu", seed=1)
-        Xbf = Xr.bfloat16()  # skip standardize: identity test
+        clf = _make_clf(shape)
+        Xbf = clf.stage(Xr)  # identity standardizer (never fit): pads only
         clf.train_epochs(
             Xbf, yr, epochs=3, batch_size=len(Xr), lr=1e-3, world_size=world
         )
@@ -114,22 +121,21 @@ def _tabular_worker(rank, world, port, out_dir):
         dist.destroy_process_group()
 
 
-def run_tabular_dp():
+def run_tabular_dp(shape=(64, 32, 10)):
     import torch.multiprocessing as mp
 
-    from unionml_amd.ops.tabular import TabularMLP
     from unionml_amd.parallel.launch import _free_port
 
     # single-process oracle: full batch, invBtot = 1/N
-    X, y = _make_data()
-    clf = TabularMLP(device="cpu", seed=1)
-    clf.train_epochs(X.bfloat16(), y, epochs=3, batch_size=N_ROWS, lr=1e-3)
+    X, y = _make_data(shape)
+    clf = _make_clf(shape)
+    clf.train_epochs(clf.stage(X), y, epochs=3, batch_size=N_ROWS, lr=1e-3)
     single_master = clf.master.clone()
 
     with tempfile.TemporaryDirectory() as out_dir:
         port = _free_port()
         mp.start_processes(
-            _tabular_worker, args=(2, port, out_dir), nprocs=2, join=True,
+            _tabular_worker, args=(2, port, out_dir, shape), nprocs=2, join=True,
             start_method="spawn",
         )
         with open(Path(out_dir) / "dp_master.pkl", "rb") as f:

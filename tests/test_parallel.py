@@ -172,3 +172,16 @@ def test_dp_worker_failure_surfaces_clean_error():
     model = build_failing_model()
     with pytest.raises(RuntimeError, match="injected trainer failure"):
         model.train(dp=2, n=40)
+
+
+def test_tabular_dp2_matches_single_gen_shape():
+    """The GENERALIZED-geometry DP path (reduce-only step -> all_reduce
+    -> Adam) over gloo must equal single-process training on the
+    concatenated data — same semantics the gen HIP kernels follow on a
+    GPU node (MNIST-like padded shape)."""
+    from dp_app import run_tabular_dp
+
+    single_master, dp_master = run_tabular_dp(shape=(100, 50, 7))
+    assert torch.allclose(single_master, dp_master, rtol=1e-4, atol=1e-6), (
+        (single_master - dp_master).abs().max()
+    )
