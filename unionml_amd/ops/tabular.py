@@ -116,7 +116,12 @@ class TabularMLP:
         big, small = self._GEN_RTS[self.g.hid]
         if self.g.cpad > 16:
             return big  # 32-class head compiles only at the default rows/WG
-        if batch is not None and (batch + big - 1) // big < 16:
+        # smaller rows/WG doubles the grid (the bandwidth-bound fwd/bwd
+        # scales with resident CUs: measured -12% at B=2048, -11% at
+        # B=4096 on the MNIST shape) — but every extra WG re-reads the
+        # whole W1 image and adds a full-nparam gradient slab, which
+        # turns pathological past ~128 WGs (B=8192: 335 vs 65 us).
+        if batch is not None and (batch + small - 1) // small <= 128:
             return small
         return big
 
