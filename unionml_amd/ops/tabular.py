@@ -28,15 +28,14 @@ Python): tests/integration/pytorch_app/quickstart.py:14-70.
 """
 
 import math
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional
 
 import torch
 
 from unionml_amd._logging import logger
-from unionml_amd.ops import hip_available, hip_ext
+from unionml_amd.ops import hip_ext
 from unionml_amd.ops import reference as ref
-from unionml_amd.ops.reference import CLS, CPAD, HID, IN, NPARAM, OFF_B1, OFF_B2, OFF_W1, OFF_W2
-from unionml_amd.ops.reference import Geometry
+from unionml_amd.ops.reference import CPAD, HID, IN, Geometry
 
 ADAM_BETA1, ADAM_BETA2, ADAM_EPS = 0.9, 0.999, 1e-8
 
