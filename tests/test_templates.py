@@ -48,6 +48,7 @@ def test_basic_template_main_trains_and_saves(tmp_path, monkeypatch):
     "template,module_ok",
     [
         ("pytorch-mi355x", True),
+        ("mnist-mi355x", True),
         ("basic-aws-lambda", True),
         ("basic-aws-lambda-s3", True),
         ("resnet-dp", True),
@@ -67,3 +68,19 @@ def test_templates_import_clean(tmp_path, monkeypatch, template, module_ok):
         env=_env(),
     )
     assert (proc.returncode == 0) == module_ok, proc.stderr[-3000:]
+
+
+@pytest.mark.timeout(300)
+def test_mnist_template_trains(tmp_path, monkeypatch):
+    """The generalized-geometry template trains end to end on CPU."""
+    monkeypatch.chdir(tmp_path)
+    result = runner.invoke(app, ["init", "mnist_app", "--template", "mnist-mi355x"])
+    assert result.exit_code == 0, result.output
+    proc = subprocess.run(
+        [sys.executable, "-c",
+         "from app import model; obj, m = model.train(trainer_kwargs={'epochs': 3}, n=600); "
+         "assert m['test'] > 0.8, m; assert obj.g.in_features == 784"],
+        cwd=tmp_path / "mnist_app", capture_output=True, text=True, timeout=240,
+        env=_env(),
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
