@@ -99,3 +99,60 @@ def test_cron_next_fire_matches_and_is_future(when):
     while probe < nxt.replace(second=0, microsecond=0):
         assert not cron_matches("*/5 * * * *", probe)
         probe += datetime.timedelta(minutes=1)
+
+
+# ---------------- generalized-geometry invariants (r02) ----------------
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    inf=st.integers(min_value=1, max_value=2048),
+    hid=st.integers(min_value=1, max_value=256),
+    cls=st.integers(min_value=2, max_value=32),
+)
+def test_geometry_invariants(inf, hid, cls):
+    """Padding math holds for every constructible geometry: alignment,
+    monotone containment, offset chain consistency, slab bounds."""
+    from unionml_amd.ops.reference import Geometry
+
+    g = Geometry(inf, hid, cls)
+    assert g.inp % 32 == 0 and g.inp >= inf and g.inp - inf < 32
+    assert g.hid >= hid and g.hid in (32, 64, 128, 256)
+    assert g.cpad in (16, 32) and g.cpad >= cls
+    assert g.off_b1 == g.inp * g.hid
+    assert g.off_w2 == g.off_b1 + g.hid
+    assert g.off_b2 == g.off_w2 + g.hid * g.cpad
+    assert g.nparam == g.off_b2 + g.cpad
+    assert g.slab_stride % 16 == 0
+    assert g.wimg_n == g.hid * g.inp + g.hid * 32 + g.cpad * g.hid
+    if g.is_specialized:
+        assert (inf, hid, cls) == (64, 32, 10)
+
+
+@settings(max_examples=50, deadline=None)
+@given(
+    inf=st.integers(min_value=2, max_value=300),
+    hid=st.integers(min_value=2, max_value=256),
+    cls=st.integers(min_value=2, max_value=32),
+    n=st.integers(min_value=1, max_value=64),
+)
+def test_cpu_step_padded_params_stay_zero(inf, hid, cls, n):
+    """One CPU training step at a random geometry never leaks gradient
+    or parameter mass into the padded region."""
+    import torch
+
+    from unionml_amd.ops import reference as ref
+    from unionml_amd.ops.tabular import TabularMLP
+
+    clf = TabularMLP(in_features=inf, hidden=hid, classes=cls, device="cpu", seed=0)
+    X = torch.randn(n, inf)
+    y = torch.randint(0, cls, (n,), dtype=torch.int32)
+    clf.fit_standardizer(X)
+    clf.train_epochs(clf.stage(X), y, epochs=1, batch_size=n, lr=1e-3)
+    W1, b1, W2, b2 = ref.unpack_master_g(clf.g, clf.master)
+    assert (W1[inf:, :] == 0).all() and (W1[:, hid:] == 0).all()
+    assert (W2[hid:, :] == 0).all() and (W2[:, cls:] == 0).all()
+    assert (b1[hid:] == 0).all() and (b2[cls:] == 0).all()
