@@ -391,6 +391,12 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
   }
   __syncthreads();
 
+  // prefetch dW1's FIRST transposed-X tile now: xbuf is free (fwd1's
+  // last Xs read retired at the barrier above) and nothing between
+  // here and dW1 touches it, so the load's round trip hides under the
+  // fwd2/dH/dW2 phases
+  load_xt_tile(L, 0, Xbf, inp, row0, (long long)B, 0, min(G::TK, inp));
+
   // ---- fwd2: L^T = W2T @ B(Hs); wave-shuffle softmax + xent over NCT
   // class tiles (classes <= 16: one MFMA tile; 17..32: two tiles, the
   // max/sum folds across the register-resident tile axis first, then
@@ -527,9 +533,8 @@ mlp_step_gen_kernel(const u16* __restrict__ Xbf,   // [N][inp] staged bf16 (zero
   }
 
   // ---- dW1^T = DHT @ B(XT), K'-streamed over inp -> slab -------------------
-  // (xbuf is free: fwd1's last Xs read completed before its loop-end
-  // barrier; dW2 above touches only HT/DLT)
-  load_xt_tile(L, 0, Xbf, inp, row0, (long long)B, 0, min(G::TK, inp));
+  // (tile 0 was prefetched before fwd2; the barrier below also covers
+  // the dH-phase atomics)
   __syncthreads();
   for (int kt = 0; kt < nkt; ++kt) {
     const int cur = G::DB ? (kt & 1) : 0;
