@@ -133,6 +133,16 @@ def train_data_parallel(
         # already inside a torchrun-style launch: act as our own rank
         return run_sharded_train_body(model, **kwargs)
 
+    # RCCL needs one distinct device per rank: dp > visible GPUs would
+    # die deep inside communicator init with an opaque error — fail
+    # here with an actionable one instead (CPU/gloo has no such limit)
+    if torch.cuda.is_available() and dp > torch.cuda.device_count():
+        raise ValueError(
+            f"dp={dp} exceeds the {torch.cuda.device_count()} visible GPU(s): "
+            "RCCL requires one device per rank. Reduce dp or widen "
+            "HIP_VISIBLE_DEVICES."
+        )
+
     import cloudpickle
 
     model_blob = cloudpickle.dumps(model)
