@@ -105,3 +105,19 @@ def test_backend_gpu_execution_and_fanout(tmp_path, monkeypatch):
         assert len(model.remote_list_prediction_ids()) >= 3
     finally:
         sys.modules.pop("gpu_backend_app", None)
+
+
+@pytest.mark.gpu
+def test_dp_oversubscription_raises_clearly():
+    """dp > visible GPUs must fail fast with an actionable error, not
+    die inside RCCL communicator init (only meaningful on boxes with
+    fewer GPUs than the requested dp)."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    n = torch.cuda.device_count()
+    from unionml_amd.models.mlp import model
+
+    with pytest.raises(ValueError, match="exceeds"):
+        model.train(dp=n + 1, synthetic=True, n=256, trainer_kwargs={"epochs": 1})
