@@ -77,6 +77,11 @@ class PinnedStager:
             self._inflight[t.dtype] = ev
         if self.copy_stream is not None:
             torch.cuda.current_stream(self.device).wait_stream(self.copy_stream)
+            # the tensor was allocated on the copy stream but lives on
+            # the compute stream from here: tell the caching allocator,
+            # or a later free could recycle the block while compute
+            # kernels still read it
+            dev.record_stream(torch.cuda.current_stream(self.device))
         return dev
 
 
