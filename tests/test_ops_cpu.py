@@ -126,3 +126,29 @@ def test_tabular_mlp_pickle_roundtrip():
     for k, v in clf.state_dict().items():
         assert torch.equal(v, back.state_dict()[k]), k
     assert back.slabs is None and back._graph is None
+
+
+def test_missing_extension_is_loud_on_gpu(monkeypatch):
+    """On a machine that claims a GPU, a missing/unloadable HIP
+    extension must raise KernelExtensionNotBuilt — the HIP path never
+    silently falls back to eager torch there (driver contract: GPU
+    tests passing on a silent fallback read as 'native code not
+    loaded')."""
+    import torch
+
+    import unionml_amd.ops as ops
+    from unionml_amd.exceptions import KernelExtensionNotBuilt
+
+    monkeypatch.setattr(ops, "_ext", None)
+    monkeypatch.setattr(ops, "_load_error", ImportError("no .so for this arch"))
+    monkeypatch.setattr(ops, "_try_load", lambda: None)
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+
+    with pytest.raises(KernelExtensionNotBuilt, match="build_ext"):
+        ops.hip_ext()  # required=None resolves to GPU-present -> required
+
+    with pytest.raises(KernelExtensionNotBuilt):
+        ops.hip_ext(required=True)
+
+    # explicit CPU-style call still returns None quietly
+    assert ops.hip_ext(required=False) is None
