@@ -75,6 +75,9 @@ def parse_args():
                    help="concurrent HTTP clients in serve mode")
     p.add_argument("--serve-workers", type=int, default=1,
                    help="SO_REUSEPORT server workers in serve mode")
+    p.add_argument("--serve-app", choices=["digits", "mnist"], default="digits",
+                   help="serve mode: flagship digits app (specialized kernels) "
+                   "or the MNIST-shape app (generalized kernels)")
     return p.parse_args()
 
 
@@ -171,13 +174,23 @@ def run_serve_mode(args):
         s.bind(("127.0.0.1", 0))
         port = s.getsockname()[1]
 
-    # train a quick artifact for the server to load (synthetic digits)
+    # train a quick artifact for the server to load (synthetic data)
     import tempfile
 
-    from unionml_amd.models.mlp import model as train_model
+    if args.serve_app == "mnist":
+        from unionml_amd.models.mnist import model as train_model
 
-    train_model.artifact = None
-    train_model.train(trainer_kwargs={"epochs": 5, "lr": 3e-3}, synthetic=True, n=4096)
+        train_model.artifact = None
+        train_model.train(trainer_kwargs={"epochs": 5, "lr": 3e-3}, n=4096)
+        serve_spec = "unionml_amd.models.mnist_serve:app"
+        rng_rows = [[float((i * 7) % 16) for i in range(784)]]
+    else:
+        from unionml_amd.models.mlp import model as train_model
+
+        train_model.artifact = None
+        train_model.train(trainer_kwargs={"epochs": 5, "lr": 3e-3}, synthetic=True, n=4096)
+        serve_spec = "unionml_amd.models.mlp_serve:app"
+        rng_rows = [{f"p{i}": float((i * 7) % 16) for i in range(64)}]
     artifact_path = os.path.join(tempfile.mkdtemp(prefix="unionml_bench_"), "model.pt")
     train_model.save(artifact_path)
 
@@ -187,15 +200,15 @@ def run_serve_mode(args):
         # the framework's SO_REUSEPORT multi-worker supervisor (each
         # worker owns its bucketed inference hipGraphs)
         server = subprocess.Popen(
-            [sys.executable, "-m", "unionml_amd.cli", "serve",
-             "unionml_amd.models.mlp_serve:app", "--host", "127.0.0.1",
+            [sys.executable, "-m", "unionml_amd.cli", "serve", serve_spec,
+             "--host", "127.0.0.1",
              "--port", str(port), "--workers", str(args.serve_workers)],
             env=env,
         )
     else:
         server = subprocess.Popen(
             [sys.executable, "-m", "uvicorn", "--host", "127.0.0.1", "--port",
-             str(port), "--log-level", "warning", "unionml_amd.models.mlp_serve:app"],
+             str(port), "--log-level", "warning", serve_spec],
             env=env,
         )
     try:
@@ -203,7 +216,6 @@ def run_serve_mode(args):
         # artifact and captures its own hipGraphs independently, so
         # demand a long streak of consecutive healthy responses (enough
         # to have hit every worker with high probability)
-        rng_rows = [{f"p{i}": float((i * 7) % 16) for i in range(64)}]
         body = json.dumps({"features": rng_rows}).encode()
         deadline = time.time() + 180
         streak, need = 0, 16 * max(1, args.serve_workers)
@@ -273,7 +285,8 @@ def run_serve_mode(args):
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "digits_mlp_64x32x10",
+                "model": "mnist_mlp_784x128x10" if args.serve_app == "mnist"
+                else "digits_mlp_64x32x10",
                 "global_batch": 1,
                 "seq_len": None,
                 "parallelism": "serve1",
