@@ -18,13 +18,19 @@
 //     layouts (W1T row-major over K, W2s K-padded, W2T) kept in sync
 //     by the in-kernel Adam phase — the same packed-image scheme the
 //     specialized kernel's wimg uses.
-//   * Everything else keeps the proven structure: one fused launch =
-//     fwd + bwd + per-WG grad slab (plain stores) + agent-scope
-//     release/acquire epoch-tag handshake (guide §6 G16) + cross-WG
-//     reduce + Adam. MFMA __builtin_amdgcn_mfma_f32_16x16x32_bf16
-//     with the hardware-verified fragment mapping documented in
-//     tabular_kernels.hip:34-46. Classes are runtime (<= 16, one MFMA
-//     tile): the wave-shuffle softmax masks c >= cls.
+//   * The training step is TWO kernels: the slab-producing MFMA
+//     fwd/bwd (grid = batch rows / rows-per-WG, plain per-WG stores,
+//     no atomics) and a wide-grid reduce+Adam kernel whose launch
+//     boundary doubles as the inter-workgroup barrier — at generalized
+//     nparam sizes an in-grid reduce is bandwidth-starved (memory
+//     parallelism scales with resident CUs), so the specialized
+//     kernel's G16 epoch-tag handshake is deliberately NOT used here
+//     (A/B trail: profiles/r02_gen_kernel_stats.md). MFMA
+//     __builtin_amdgcn_mfma_f32_16x16x32_bf16 with the
+//     hardware-verified fragment mapping documented in
+//     tabular_kernels.hip:34-46. Classes are runtime (<= 32, one or
+//     two MFMA tiles): the wave-shuffle softmax masks c >= cls and
+//     folds across the register-resident class-tile axis.
 //
 // All launches are stream-ordered and hipGraph-capturable; the Adam
 // step counter lives in device memory so bias correction is exact
