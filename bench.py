@@ -268,6 +268,8 @@ def run_serve_mode(args):
             elapsed = time.perf_counter() - t_start
         errors = sum(1 for x in lats if x != x)
         lats = [x for x in lats if x == x]
+        if not lats:
+            raise RuntimeError(f"serve mode: all {errors} requests failed")
         steps = len(lats)
         lats.sort()
         p = lambda q: lats[min(len(lats) - 1, int(q * len(lats)))]  # noqa: E731
