@@ -413,3 +413,68 @@ def test_reload_token_gate(tmp_path, monkeypatch):
         assert client.post("/reload").status_code == 403
         assert client.post("/reload", headers={"X-Reload-Token": "wrong"}).status_code == 403
         assert client.post("/reload", headers={"X-Reload-Token": "s3cret"}).status_code == 200
+
+
+@pytest.mark.timeout(240)
+def test_serve_multiworker_supervisor(tmp_path):
+    """`unionml-amd serve --workers 2`: the SO_REUSEPORT supervisor
+    boots N single-worker processes sharing the port; requests succeed
+    (kernel load-balances across workers), and SIGTERM tears the whole
+    tree down cleanly."""
+    import os
+    import signal
+
+    import httpx
+
+    import unionml_amd
+
+    env = dict(
+        os.environ,
+        PYTHONPATH=str(Path(unionml_amd.__file__).parent.parent)
+        + os.pathsep
+        + os.environ.get("PYTHONPATH", ""),
+    )
+    (tmp_path / "serve_app.py").write_text(SERVE_APP)
+    proc = subprocess.run(
+        [sys.executable, "serve_app.py"], cwd=tmp_path, capture_output=True, text=True, env=env
+    )
+    assert proc.returncode == 0, proc.stderr
+
+    port = _free_port()
+    server = subprocess.Popen(
+        [sys.executable, "-m", "unionml_amd.cli", "serve", "serve_app:app",
+         "--model-path", "model.joblib", "--port", str(port), "--workers", "2"],
+        cwd=tmp_path, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, env=env,
+    )
+    try:
+        deadline = time.monotonic() + 90
+        streak = 0
+        while time.monotonic() < deadline and streak < 8:  # hit both workers
+            if server.poll() is not None:
+                raise AssertionError(
+                    f"supervisor exited early:\n{server.stdout.read().decode()[-3000:]}"
+                )
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/health", timeout=1.0).status_code == 200:
+                    streak += 1
+                    continue
+            except httpx.HTTPError:
+                pass
+            streak = 0
+            time.sleep(0.5)
+        assert streak >= 8, "workers never became (all) healthy"
+
+        for _ in range(10):  # round-robins across both workers
+            resp = httpx.post(
+                f"http://127.0.0.1:{port}/predict",
+                json={"features": [{"x1": 0.9, "x2": 0.9, "x3": 0.9}]},
+                timeout=10.0,
+            )
+            assert resp.status_code == 200, resp.text
+
+        server.send_signal(signal.SIGTERM)
+        server.wait(timeout=20)   # supervisor reaps its workers
+        assert server.returncode is not None
+    finally:
+        if server.poll() is None:
+            server.kill()
