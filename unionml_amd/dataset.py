@@ -9,7 +9,7 @@ SQL constructors, and a compiled ``dataset_task``.
 MI355X-native additions:
 
 - :meth:`Dataset.stage_to_device` — stages a parsed split into a GPU's
-  HBM through a pinned host buffer + async H2D copy on a side stream
+  HBM via async H2D copies on a side stream
   (SURVEY.md §2c "pinned staging loader").
 - The corrected default-parser column logic (the reference inverts its
   feature-selection guard — SURVEY.md §8 "Known reference quirks").
@@ -319,7 +319,7 @@ class Dataset(TrackedInstance):
         """Stage a parsed split's arrays into GPU HBM.
 
         Converts DataFrame/ndarray elements to torch tensors, moves them
-        through a pinned host buffer and issues ``hipMemcpyAsync`` H2D on
+        through the pinned transfer pool and issues ``hipMemcpyAsync`` H2D on
         ``stream`` (or the current stream). Non-array elements pass through.
         """
         from unionml_amd.utils.staging import stage_split_to_device

@@ -1,21 +1,32 @@
-"""Pinned-buffer H2D staging for parsed splits (SURVEY.md §2c row
-'pinned staging loader').
+"""H2D staging for parsed splits (SURVEY.md §2c row 'pinned staging
+loader').
 
 Feature matrices come out of the dataset pipeline as DataFrames or
-ndarrays; this module turns them into device tensors by way of a
-*reused* pinned host buffer and an async H2D copy on a dedicated copy
-stream, so staging overlaps compute. With 288 GB of HBM3E per MI355X
-the whole dataset is staged resident — tensors are uploaded once and
+ndarrays; this module turns them into device tensors on a dedicated
+copy stream so staging overlaps compute, and keeps them HBM-resident —
+with 288 GB of HBM3E per MI355X the whole dataset is uploaded once and
 re-read from HBM, never re-staged per epoch.
+
+Measured design note (profiles/raw_r02/staging_bw.json /
+staging_lat.json): a hand-rolled reused pinned buffer
+(host memcpy -> pinned -> hipMemcpyAsync) is STRICTLY SLOWER on this
+stack than PyTorch's own transfer engine at every size (64 KB: 41 vs
+13 µs; 1 GB: 10 vs 51 GB/s) — the single-threaded host copy into the
+pinned buffer is the bottleneck, while torch's path chunks through its
+internal pinned pool with hipMemcpyAsync. So this stager fronts
+torch's transfer engine (pinned + async under the hood) and adds the
+copy-stream overlap + allocator stream-safety that raw ``.to()`` calls
+lack.
 """
 
-from typing import Any, List, Optional
+from typing import Any, List
 
 import numpy as np
 
 
 class PinnedStager:
-    """Reusable pinned host buffer + copy stream for one device."""
+    """Copy-stream H2D staging for one device (torch transfer engine
+    underneath — see the module docstring for the measurements)."""
 
     def __init__(self, device=None):
         import torch
@@ -24,29 +35,13 @@ class PinnedStager:
         self.device = torch.device(device) if device is not None else torch.device(
             "cuda" if torch.cuda.is_available() else "cpu"
         )
-        self._pinned: dict = {}  # dtype -> host buffer, grown geometrically
-        self._inflight: dict = {}  # dtype -> event of the last async copy FROM the buffer
         self.copy_stream = (
             torch.cuda.Stream(device=self.device) if self.device.type == "cuda" else None
         )
 
-    def _pinned_buffer(self, numel: int, dtype):
-        # the previous async H2D reading this buffer must have completed
-        # before the host overwrites it (wait_stream orders GPU streams,
-        # not host writes — without this, back-to-back stagings race)
-        ev = self._inflight.pop(dtype, None)
-        if ev is not None:
-            ev.synchronize()
-        buf = self._pinned.get(dtype)
-        if buf is None or buf.numel() < numel:
-            cap = max(numel, 2 * buf.numel() if buf is not None else numel)
-            buf = self.torch.empty(cap, dtype=dtype, pin_memory=self.device.type == "cuda")
-            self._pinned[dtype] = buf
-        return buf
-
     def to_device(self, array: Any, non_blocking: bool = True):
-        """array/DataFrame/tensor -> tensor on the target device via the
-        pinned buffer (hipMemcpyAsync under the hood on ROCm)."""
+        """array/DataFrame/tensor -> tensor on the target device
+        (hipMemcpyAsync through torch's pinned staging pool)."""
         torch = self.torch
         if hasattr(array, "to_numpy"):  # DataFrame / Series
             array = array.to_numpy()
@@ -61,26 +56,17 @@ class PinnedStager:
 
         if self.device.type != "cuda":
             return t.clone()
-
         if t.device.type == "cuda":
             return t.to(self.device)
 
-        flat = t.reshape(-1)
-        host = self._pinned_buffer(flat.numel(), t.dtype)[: flat.numel()]
-        host.copy_(flat)
         stream = self.copy_stream or torch.cuda.current_stream(self.device)
         with torch.cuda.stream(stream):
-            dev = torch.empty(t.shape, dtype=t.dtype, device=self.device)
-            dev.reshape(-1).copy_(host, non_blocking=non_blocking)
-            ev = torch.cuda.Event()
-            ev.record(stream)
-            self._inflight[t.dtype] = ev
+            dev = t.to(self.device, non_blocking=non_blocking)
         if self.copy_stream is not None:
             torch.cuda.current_stream(self.device).wait_stream(self.copy_stream)
-            # the tensor was allocated on the copy stream but lives on
-            # the compute stream from here: tell the caching allocator,
-            # or a later free could recycle the block while compute
-            # kernels still read it
+            # allocated on the copy stream, consumed on the compute
+            # stream: tell the caching allocator or a later free could
+            # recycle the block under in-flight reads
             dev.record_stream(torch.cuda.current_stream(self.device))
         return dev
 
