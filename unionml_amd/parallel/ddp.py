@@ -18,6 +18,12 @@ framework's own reducer, not ``torch.nn.parallel.DDP``:
   all-reduce at half the bytes of fp32).
 
 Works identically over ``gloo`` for CPU-only multi-process tests.
+
+Known limitation (same class of issue torch DDP's
+``find_unused_parameters`` exists for): every rank must run the same
+model graph each step. If a parameter receives a gradient on some ranks
+but not others (rank-divergent control flow), the per-bucket collectives
+mismatch and the job can deadlock — shard DATA, not model structure.
 """
 
 import os
