@@ -274,6 +274,8 @@ def test_torch_default_saver_loader_roundtrip(tmp_path):
 
     ds = Dataset(name="torch_art", targets=["y"], test_size=0.25, random_state=0)
 
+    import numpy as np
+
     @ds.reader
     def reader(n: int = 80) -> pd.DataFrame:
         rng = np.random.RandomState(1)
