@@ -40,6 +40,27 @@ app = typer.Typer(
 )
 
 
+def _version_callback(value: bool):
+    if value:
+        try:
+            from importlib.metadata import version as _v
+
+            typer.echo(f"unionml-amd {_v('unionml_amd')}")
+        except Exception:
+            typer.echo("unionml-amd 0.1.0")
+        raise typer.Exit()
+
+
+@app.callback()
+def _main(
+    version: bool = typer.Option(
+        False, "--version", callback=_version_callback, is_eager=True,
+        help="print the framework version and exit",
+    ),
+):
+    pass
+
+
 @app.callback(invoke_without_command=True)
 def _main_callback(
     version: bool = typer.Option(False, "--version", help="print version and exit"),
