@@ -72,3 +72,18 @@ def test_lambda_handler_for_falls_back_without_mangum():
         assert isinstance(h, MiniAsgiLambdaAdapter)
     resp = h(make_api_gateway_event("/health"), None)
     assert resp["statusCode"] == 200
+
+
+def test_base64_encoded_event_body(handler):
+    """API Gateway may deliver base64-encoded bodies; the adapter must
+    decode them before handing the bytes to the ASGI app."""
+    import base64
+
+    feats = [{"x1": 0.5, "x2": 0.1, "x3": 0.9}]
+    raw = json.dumps({"features": feats}).encode()
+    event = make_api_gateway_event("/predict", method="POST")
+    event["body"] = base64.b64encode(raw).decode()
+    event["isBase64Encoded"] = True
+    resp = handler(event, context=None)
+    assert resp["statusCode"] == 200, resp
+    assert len(json.loads(resp["body"])) == 1
