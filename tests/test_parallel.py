@@ -164,6 +164,64 @@ def test_reducer_broadcasts_initial_params(tmp_path):
     assert (tmp_path / "ok_0").exists() and (tmp_path / "ok_1").exists()
 
 
+def _partial_bucket_worker(rank, world, port, out_dir):
+    """A parameter that produces no grad this step (params-present-but-
+    unused) leaves its bucket partial; finalize() must fall back to the
+    synchronous reduce instead of hanging or skipping the used grads."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+
+        class Partial(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.used = torch.nn.Linear(8, 4)
+                self.unused = torch.nn.Linear(8, 4)  # never in forward
+
+            def forward(self, x):
+                return self.used(x)
+
+        model = Partial()
+        # one big bucket so used + unused params share it -> partial
+        reducer = GradientAllReducer(model, bucket_mb=64)
+        assert len(reducer.buckets) == 1
+        torch.manual_seed(200 + rank)
+        x = torch.randn(16, 8)
+        y = torch.randint(0, 4, (16,))
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        loss.backward()
+        b = reducer.buckets[0]
+        assert 0 < b.pending < len(b.params), "test premise: bucket partial"
+        reducer.finalize()
+
+        # every rank must hold the same averaged grads afterwards
+        grads = torch.cat([p.grad.reshape(-1) for p in model.parameters()])
+        gathered = [torch.zeros_like(grads) for _ in range(world)]
+        dist.all_gather(gathered, grads)
+        for g in gathered:
+            assert torch.allclose(g, grads, atol=1e-6), "ranks disagree"
+        # unused params: zero grad (averaged zeros), used params: nonzero
+        assert model.unused.weight.grad.abs().max() == 0
+        assert model.used.weight.grad.abs().max() > 0
+        with open(os.path.join(out_dir, f"ok_{rank}"), "w") as f:
+            f.write("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_reducer_partial_bucket_world2(tmp_path):
+    from unionml_amd.parallel.launch import _free_port
+
+    port = _free_port()
+    mp.start_processes(
+        _partial_bucket_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True,
+        start_method="spawn",
+    )
+    assert (tmp_path / "ok_0").exists() and (tmp_path / "ok_1").exists()
+
+
 def test_dp_worker_failure_surfaces_clean_error():
     """A trainer crashing inside a DP worker must surface as a
     RuntimeError with the worker traceback, not a hang."""
