@@ -35,7 +35,9 @@ def tensors_to_cpu(obj: Any, _copy_modules: bool = True) -> Any:
     if torch.is_tensor(obj):
         return obj.detach().cpu() if obj.is_cuda else obj
     if isinstance(obj, tuple):
-        return tuple(tensors_to_cpu(o, _copy_modules) for o in obj)
+        vals = [tensors_to_cpu(o, _copy_modules) for o in obj]
+        # preserve NamedTuple subtypes (e.g. splitter outputs)
+        return type(obj)(*vals) if hasattr(obj, "_fields") else tuple(vals)
     if isinstance(obj, list):
         return [tensors_to_cpu(o, _copy_modules) for o in obj]
     if isinstance(obj, dict):
