@@ -159,3 +159,49 @@ def test_cpu_train_predict_wide_classes(shape):
     preds = clf.predict(X)
     acc = (preds == y).float().mean().item()
     assert acc > 0.9, f"{shape}: accuracy {acc}, loss {loss}"
+
+
+def test_checkpoint_resumes_training_exactly():
+    """Full checkpoints carry Adam state: 2+2 epochs through a
+    state_dict roundtrip must be bit-identical to 4 straight epochs
+    (same deterministic batch order, same moments, same step count)."""
+    torch.manual_seed(11)
+    X = torch.randn(128, 100) * 3
+    y = torch.randint(0, 7, (128,), dtype=torch.int32)
+
+    a = TabularMLP(in_features=100, hidden=50, classes=7, device="cpu", seed=0)
+    a.fit_standardizer(X)
+    a.train_epochs(a.stage(X), y, epochs=4, batch_size=32, lr=1e-3)
+
+    b = TabularMLP(in_features=100, hidden=50, classes=7, device="cpu", seed=0)
+    b.fit_standardizer(X)
+    b.train_epochs(b.stage(X), y, epochs=2, batch_size=32, lr=1e-3)
+    state = b.state_dict()
+
+    c = TabularMLP(in_features=100, hidden=50, classes=7, device="cpu", seed=99)
+    c.load_state_dict(state)  # restores weights, standardizer, m/v/t
+    c.train_epochs(c.stage(X), y, epochs=2, batch_size=32, lr=1e-3)
+
+    assert int(c.t_dev.item()) == int(a.t_dev.item()) == 16
+    assert torch.equal(a.master, c.master), (
+        (a.master - c.master).abs().max()
+    )
+
+
+def test_weights_only_checkpoint_resets_optimizer():
+    """Checkpoints from before optimizer-state support (no adam_* keys)
+    still load — with fresh moments and t=0."""
+    clf = TabularMLP(in_features=100, hidden=50, classes=7, device="cpu", seed=0)
+    X = torch.randn(64, 100)
+    y = torch.randint(0, 7, (64,), dtype=torch.int32)
+    clf.fit_standardizer(X)
+    clf.train_epochs(clf.stage(X), y, epochs=2, batch_size=32, lr=1e-3)
+    state = clf.state_dict()
+    for k in ("adam_m", "adam_v", "adam_t"):
+        state.pop(k)
+
+    fresh = TabularMLP(in_features=100, hidden=50, classes=7, device="cpu")
+    fresh.load_state_dict(state)
+    assert int(fresh.t_dev.item()) == 0
+    assert fresh.m.abs().max() == 0 and fresh.v.abs().max() == 0
+    assert torch.equal(fresh.predict(X), clf.predict(X))

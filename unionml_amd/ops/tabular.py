@@ -490,7 +490,15 @@ class TabularMLP:
         self._build_wimg()
 
     def state_dict(self) -> Dict[str, torch.Tensor]:
-        """Logical (unpadded) weights + standardizer state."""
+        """Logical (unpadded) weights + standardizer + Adam state.
+
+        Includes the optimizer moments and step counter so a reloaded
+        model RESUMES training exactly (bit-identical step sequence),
+        not just serves inference. The moments are stored in the padded
+        master layout — safe because load_state_dict enforces an exact
+        geometry match, and the padded lanes are exact zeros by the
+        Adam-from-zero-init invariant.
+        """
         g = self.g
         W1, b1, W2, b2 = ref.unpack_master_g(g, self.master.cpu())
         return {
@@ -501,6 +509,9 @@ class TabularMLP:
             "mean": self.mean.cpu().clone(),
             "invstd": self.invstd.cpu().clone(),
             "geometry": torch.tensor([g.in_features, g.hidden, g.classes]),
+            "adam_m": self.m.cpu().clone(),
+            "adam_v": self.v.cpu().clone(),
+            "adam_t": self.t_dev.cpu().clone(),
         }
 
     def load_state_dict(self, state: Dict[str, torch.Tensor]):
@@ -520,5 +531,13 @@ class TabularMLP:
         self.bfmirror.copy_(self.master.bfloat16())
         self.mean.copy_(state["mean"].to(self.device))
         self.invstd.copy_(state["invstd"].to(self.device))
+        if "adam_m" in state:  # full checkpoint: resume training exactly
+            self.m.copy_(state["adam_m"].to(self.device))
+            self.v.copy_(state["adam_v"].to(self.device))
+            self.t_dev.copy_(state["adam_t"].to(self.device))
+        else:  # weights-only (pre-resume checkpoints): fresh optimizer
+            self.m.zero_()
+            self.v.zero_()
+            self.t_dev.zero_()
         self._graph = self._graph_key = None
         self._build_wimg()
