@@ -113,3 +113,47 @@ def test_scheduler_fires_training_and_prediction(sched_app):
             ex.status,
             (Path(ex.path) / "worker.log").read_text()[-2000:],
         )
+
+
+@pytest.mark.timeout(60)
+def test_scheduler_routes_workflow_level_inputs(sched_app, monkeypatch):
+    """Schedule inputs may mix workflow-level stage kwargs
+    (hyperparameters, trainer_kwargs, ...) with reader kwargs; the
+    scheduler must route each to its own workflow input instead of
+    shoving everything into reader_kwargs."""
+    import sched_app as appmod
+
+    model = appmod.model
+    backend = model._backend()
+    model.schedule_training(
+        "tuned",
+        fixed_rate=datetime.timedelta(minutes=1),
+        inputs={"hyperparameters": {"C": 0.5}, "trainer_kwargs": None, "n": 25},
+    )
+    backend.deploy(model, allow_uncommitted=True)
+
+    fired = []
+
+    def record(model, workflow, app_version=None, inputs=None, schedule_name=None, **kw):
+        fired.append((workflow, schedule_name, inputs))
+
+    monkeypatch.setattr(backend, "execute", record)
+    # minute ticks away from an hour boundary: only the fixed-rate
+    # schedules fire (the mocked execute would break the prediction
+    # schedule's artifact lookup)
+    t0 = datetime.datetime(2026, 9, 13, 11, 10, 30)
+    ticks = [t0 + datetime.timedelta(minutes=i) for i in range(4)]
+    it = iter(ticks)
+    backend.run_scheduler(model, iterations=len(ticks), poll_s=0.0, now_fn=lambda: next(it))
+
+    tuned = [f for f in fired if f[1] == "tuned"]
+    assert tuned, f"tuned schedule never fired: {fired}"
+    workflow, _, inputs = tuned[0]
+    assert workflow == "train"
+    assert inputs["hyperparameters"] == {"C": 0.5}
+    assert inputs["trainer_kwargs"] is None
+    assert inputs["reader_kwargs"] == {"n": 25}
+
+    plain = [f for f in fired if f[1] == "every2min"]
+    assert plain and plain[0][2]["reader_kwargs"] == {"n": 30}
+    assert plain[0][2]["hyperparameters"] is None

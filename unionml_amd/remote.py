@@ -579,14 +579,22 @@ class Backend:
                         inputs[lp["time_arg"]] = now
                     wf = lp["workflow"]
                     if wf == "train":
-                        exec_inputs = dict(
-                            hyperparameters=None,
-                            loader_kwargs=None,
-                            splitter_kwargs=None,
-                            parser_kwargs=None,
-                            trainer_kwargs=None,
-                            reader_kwargs=inputs,
-                        )
+                        # schedule inputs may carry workflow-level stage
+                        # kwargs (hyperparameters, trainer_kwargs, ...);
+                        # everything else is a reader kwarg (incl. the
+                        # injected time_arg)
+                        exec_inputs = {
+                            k: inputs.pop(k, None)
+                            for k in (
+                                "hyperparameters",
+                                "loader_kwargs",
+                                "splitter_kwargs",
+                                "parser_kwargs",
+                                "trainer_kwargs",
+                            )
+                        }
+                        explicit = inputs.pop("reader_kwargs", None)
+                        exec_inputs["reader_kwargs"] = {**inputs, **(explicit or {})}
                     else:
                         model_obj = inputs.pop("model_object", None)
                         if model_obj is None:
