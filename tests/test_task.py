@@ -281,3 +281,31 @@ def test_workflow_dependency_order_not_insertion_order():
     wf.add_node(Task(lambda p, q: p + q, "join"), {"p": ("node", l, None), "q": ("node", r, None)})
     assert wf(a=1) == 2 * 2 + 2 * 3
     assert [sorted(w) for w in wf._waves()] == [[0], [1, 2], [3]]
+
+
+def test_workflow_parallel_branch_error_propagates():
+    """A branch raising inside a parallel wave must fail the workflow
+    call with the ORIGINAL exception (no hang, no swallowed error), and
+    healthy sibling branches must not mask it."""
+    from unionml_amd.task import Task, Workflow
+
+    class Boom(RuntimeError):
+        pass
+
+    def bad(x):
+        raise Boom("branch exploded")
+
+    wf = Workflow(
+        "w_err",
+        inputs=["a"],
+        outputs=[("out", ("node", 2, None))],
+        parallel=True,
+    )
+    n0 = wf.add_node(Task(lambda x: x + 1, "ok"), {"x": ("input", "a")})
+    n1 = wf.add_node(Task(bad, "bad"), {"x": ("input", "a")})
+    wf.add_node(
+        Task(lambda l, r: (l, r), "join"),
+        {"l": ("node", n0, None), "r": ("node", n1, None)},
+    )
+    with pytest.raises(Boom, match="branch exploded"):
+        wf(a=1)
