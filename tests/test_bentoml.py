@@ -138,3 +138,26 @@ def test_infer_framework_and_io_descriptor(bentoml_env):
     assert infer_io_descriptor(List[float]) == "JSON"
     assert infer_io_descriptor(Dict[str, float]) == "JSON"
     assert infer_io_descriptor(int) is None
+
+
+def test_configure_infers_io_from_app_types(bentoml_env):
+    """configure() without explicit features/predictions infers the IO
+    descriptors from the dataset's declared feature type (unwrapping
+    FeatureTypeUnion to the serve-time side) and the predictor's return
+    annotation (reference defaults its IO mapping the same way:
+    services/bentoml.py:33-38, 238-247)."""
+    from unionml_amd.services.bentoml import BentoMLService
+
+    model = build_sklearn_app()
+
+    @model._dataset.feature_loader
+    def feature_loader(data) -> np.ndarray:
+        return np.asarray(data, dtype=np.float64)
+
+    model.train()
+    svc = BentoMLService(model, name="io_infer_app").configure()
+    api = svc.apis["predict"]
+    # FeatureTypeUnion[dataset_type, np.ndarray] -> serve side -> NumpyNdarray
+    assert type(api["input"]).__name__ == "NumpyNdarray", type(api["input"])
+    # predictor returns List[float] -> JSON
+    assert type(api["output"]).__name__ == "JSON", type(api["output"])

@@ -188,6 +188,17 @@ def create_service(
     bentoml = _require_bentoml()
     import bentoml.io as bio
 
+    if features is None:  # default to the dataset's declared feature type
+        try:
+            features = service.model._dataset.feature_type
+        except Exception:
+            features = None
+    if predictions is None and service.model._predictor is not None:
+        import inspect
+
+        ra = inspect.signature(service.model._predictor).return_annotation
+        predictions = None if ra is inspect.Signature.empty else ra
+
     in_name = (features is not None and infer_io_descriptor(features)) or "JSON"
     out_name = (predictions is not None and infer_io_descriptor(predictions)) or "JSON"
     input_io = getattr(bio, in_name)()
